@@ -1,0 +1,168 @@
+"""Batched Reed-Solomon encode/decode on MI355X — the GPU mirror of the
+reference's ReedSolomon<32,32> (reed_solomon.h:41-373) plugin surface.
+
+Parts are torch uint8 tensors resident on the GPU; PyTorch provides memory
+and streams, liblizec.so provides the compute.  Semantics mirror the
+reference exactly: parts indexed 0..k+m-1 (data then parity,
+slice_traits.h:183-197), None input = implicit zero part
+(reed_solomon.h:79), exactly m erased parts per recover call
+(reed_solomon.h:95), per-(erasure-pattern) table cache
+(reed_solomon.h:194-198).
+"""
+import ctypes
+
+import numpy as np
+import torch
+
+from . import lib as L
+from . import slice_traits
+
+
+def _tables_cache_key(k, m, present, nonnull, needed):
+    return (k, m, present, nonnull, needed)
+
+
+class ReedSolomon:
+    """Reed-Solomon ec(k,m) over batches of stripes on one GPU."""
+
+    def __init__(self, k, m, device=0):
+        if not (slice_traits.MIN_DATA <= k <= slice_traits.MAX_DATA):
+            raise ValueError(f"k={k} out of [2,32]")
+        if not (slice_traits.MIN_PARITY <= m <= slice_traits.MAX_PARITY):
+            raise ValueError(f"m={m} out of [1,32]")
+        self.k = k
+        self.m = m
+        self.device = device
+        self.slice_type = slice_traits.ec_slice_type(k, m)
+        self._tables = {}
+        self._engine = L.engine(device)
+        self._lib = L.lib()
+
+    # ---------------- tables (host-side matrix algebra) ---
+
+    def _get_tables(self, present, nonnull, needed):
+        key = _tables_cache_key(self.k, self.m, present, nonnull, needed)
+        t = self._tables.get(key)
+        if t is None:
+            tbl = np.zeros(32 * 32 * 32, np.uint8)
+            ic = ctypes.c_int()
+            oc = ctypes.c_int()
+            L.check(self._lib.lizec_rs_tables(
+                self.k, self.m, present, nonnull, needed,
+                tbl.ctypes.data_as(ctypes.POINTER(ctypes.c_uint8)),
+                ctypes.byref(ic), ctypes.byref(oc)), "lizec_rs_tables")
+            t = (np.ascontiguousarray(tbl[:32 * ic.value * oc.value]),
+                 ic.value, oc.value)
+            self._tables[key] = t
+        return t
+
+    # ---------------- batch ops ----------------
+
+    def _run(self, part_len, tbl, ic, oc, src_ptrs, dst_ptrs, nstripes):
+        stream = torch.cuda.current_stream(self.device).cuda_stream
+        L.check(self._lib.lizec_ec_encode_batch(
+            self._engine, part_len, ic, oc,
+            tbl.ctypes.data_as(ctypes.POINTER(ctypes.c_uint8)),
+            src_ptrs.ctypes.data_as(ctypes.POINTER(ctypes.c_uint64)),
+            dst_ptrs.ctypes.data_as(ctypes.POINTER(ctypes.c_uint64)),
+            nstripes, ctypes.c_void_p(stream)), "lizec_ec_encode_batch")
+
+    @staticmethod
+    def _check_part(t, what):
+        if t.dtype != torch.uint8 or not t.is_cuda or not t.is_contiguous():
+            raise ValueError(f"{what} must be a contiguous CUDA uint8 tensor")
+
+    def encode_batch(self, data, parity=None):
+        """Compute parity for a batch of stripes.
+
+        data: uint8 CUDA tensor [S, k, L] (contiguous), L % 16 == 0.
+        Returns parity [S, m, L] (allocated if not given).
+        Mirrors ReedSolomon::encode (reed_solomon.h:134-155) /
+        ChunkWriter::computeParityBlock (chunk_writer.cc:365-402), batched.
+        """
+        self._check_part(data, "data")
+        S, k, plen = data.shape
+        if k != self.k:
+            raise ValueError(f"data has {k} parts, expected {self.k}")
+        if plen % 16:
+            raise ValueError("part length must be a multiple of 16")
+        if parity is None:
+            parity = torch.empty((S, self.m, plen), dtype=torch.uint8,
+                                 device=data.device)
+        else:
+            self._check_part(parity, "parity")
+
+        dmask = (1 << self.k) - 1
+        pmask = ((1 << self.m) - 1) << self.k
+        tbl, ic, oc = self._get_tables(dmask, dmask, pmask)
+        assert (ic, oc) == (self.k, self.m)
+
+        src = (data.data_ptr() +
+               np.arange(S * k, dtype=np.uint64) * np.uint64(plen))
+        dst = (parity.data_ptr() +
+               np.arange(S * self.m, dtype=np.uint64) * np.uint64(plen))
+        self._run(plen, tbl, ic, oc, src, dst, S)
+        return parity
+
+    def recover_batch(self, fragments, erased, want=None):
+        """Recover missing parts for a batch of stripes.
+
+        fragments: list of k+m entries; entry i is a uint8 CUDA tensor [S, L]
+          for an available part, or None (erased part, or available-but-zero
+          part if i not in `erased` — reed_solomon.h:79).
+        erased: iterable of part indices (exactly m of them,
+          reed_solomon.h:95; pad with available parts you don't need, as
+          ec_read_plan.h:126-133 does).
+        want: indices to reconstruct (default: all erased).
+        Returns dict part_index -> uint8 tensor [S, L].
+        """
+        nparts = self.k + self.m
+        if len(fragments) != nparts:
+            raise ValueError(f"need {nparts} fragment slots")
+        erased = frozenset(erased)
+        if len(erased) != self.m:
+            raise ValueError(f"exactly m={self.m} erased parts required "
+                             f"(got {len(erased)}); pad like ec_read_plan.h:126")
+        want = frozenset(want) if want is not None else erased
+        if not want.issubset(erased):
+            raise ValueError("want must be a subset of erased")
+
+        present = 0
+        nonnull = 0
+        needed = 0
+        S = plen = dev = None
+        for i in range(nparts):
+            if i in erased:
+                continue
+            present |= 1 << i
+            if fragments[i] is not None:
+                self._check_part(fragments[i], f"fragment {i}")
+                nonnull |= 1 << i
+                S, plen = fragments[i].shape
+                dev = fragments[i].device
+        for i in want:
+            needed |= 1 << i
+        if S is None:
+            raise ValueError("all surviving parts are None")
+        if plen % 16:
+            raise ValueError("part length must be a multiple of 16")
+
+        tbl, ic, oc = self._get_tables(present, nonnull, needed)
+
+        srcs = [fragments[i] for i in range(nparts)
+                if (nonnull >> i) & 1]
+        outs = {i: torch.empty((S, plen), dtype=torch.uint8, device=dev)
+                for i in sorted(want)}
+        stride = np.arange(S, dtype=np.uint64) * np.uint64(plen)
+        src = np.empty((S, ic), np.uint64)
+        for j, t in enumerate(srcs):
+            src[:, j] = t.data_ptr() + stride
+        dst = np.empty((S, oc), np.uint64)
+        for j, i in enumerate(sorted(want)):
+            dst[:, j] = outs[i].data_ptr() + stride
+        self._run(plen, tbl, ic, oc, np.ascontiguousarray(src.ravel()),
+                  np.ascontiguousarray(dst.ravel()), S)
+        return outs
+
+    def sync(self):
+        L.check(self._lib.lizec_engine_sync(self._engine))
