@@ -157,6 +157,18 @@ int lizec_ec_encode_batch(lizec_engine *e, uint64_t part_len, int srcs,
                           const uint64_t *src_dptrs, const uint64_t *dst_dptrs,
                           int num_stripes, void *stream);
 
+/* Prepared batches ("plans", after the reference's read-plan/matrix-cache
+ * structure, read_plan.h + reed_solomon.h:194-198): device-resident tables
+ * and pointer arrays built once, then run repeatedly with no host->device
+ * traffic.  Same arguments as lizec_ec_encode_batch. */
+typedef struct lizec_plan lizec_plan;
+int lizec_ec_plan_create(lizec_engine *e, uint64_t part_len, int srcs,
+                         int dests, const uint8_t *gftbls,
+                         const uint64_t *src_dptrs, const uint64_t *dst_dptrs,
+                         int num_stripes, lizec_plan **out);
+int lizec_ec_plan_run(lizec_plan *p, void *stream);
+void lizec_ec_plan_destroy(lizec_plan *p);
+
 /* Per-block CRC32 over a contiguous device buffer:
  *  dev_crcs_out[b] = lizec_crc32(seed, dev_buf + b*block_len, block_len)
  * The chunkserver's per-64KiB-block gate (hddspacemgr.cc:1918-1920, scrub

@@ -6,9 +6,11 @@
  *  - GF(2^8) multiply via in-register nibble LUTs built from v_perm_b32
  *    (__builtin_amdgcn_perm): the 32-byte ISA-L coefficient table
  *    (tbl[i]=c*i, tbl[16+i]=c*(i<<4)) lives in 8 VGPRs per (dest,src) pair,
- *    read once per 64-byte strip from LDS broadcast;
- *  - 16-byte vectorized HBM loads/stores, each thread owning 4x16B strips
- *    laid out so every wave instruction is a fully-coalesced 1 KiB access;
+ *    re-read per 64-byte strip as a wave-uniform LDS broadcast;
+ *  - 16-byte vectorized HBM loads/stores laid out so every wave instruction
+ *    is a fully-coalesced 1 KiB access; full tiles run a branchless
+ *    software-pipelined path (next source's strips prefetched while the
+ *    current one is accumulated), ragged tails take a guarded path;
  *  - per-64KiB-block CRC32 (hddspacemgr.cc:1918 gate) with one wave per
  *    block: 64 lanes x 1 KiB segments, slicing-by-4 LDS tables, then a
  *    shfl-based combine tree folding segment CRCs with the same GF(2)
@@ -42,7 +44,8 @@
  * 64 consecutive 16-byte lanes = one coalesced 1 KiB segment. */
 constexpr int kThreads = 256;
 constexpr int kChunks = 4;
-constexpr uint32_t kTileBytes = kThreads * 16 * kChunks;  /* 16384 */
+constexpr uint32_t kChunkBytes = kThreads * 16;           /* 4096  */
+constexpr uint32_t kTileBytes = kChunkBytes * kChunks;    /* 16384 */
 
 /* 16-entry byte LUT on 4 packed nibbles via v_perm_b32.
  * t0..t3 = table bytes [0..3],[4..7],[8..11],[12..15]; nib = 4 nibble values
@@ -67,19 +70,32 @@ __device__ __forceinline__ uint32_t gf_macc(uint32_t acc, uint32_t w,
 	return acc ^ lut16(L.x, L.y, L.z, L.w, nl) ^ lut16(H.x, H.y, H.z, H.w, nh);
 }
 
+template <int D>
+__device__ __forceinline__ void gf_macc_all(uint4 (&acc)[D][kChunks], int c,
+                                            const uint4 &w, const uint4 (&L)[D],
+                                            const uint4 (&H)[D]) {
+#pragma unroll
+	for (int d = 0; d < D; ++d) {
+		acc[d][c].x = gf_macc(acc[d][c].x, w.x, L[d], H[d]);
+		acc[d][c].y = gf_macc(acc[d][c].y, w.y, L[d], H[d]);
+		acc[d][c].z = gf_macc(acc[d][c].z, w.z, L[d], H[d]);
+		acc[d][c].w = gf_macc(acc[d][c].w, w.w, L[d], H[d]);
+	}
+}
+
 /* D = number of destination parts computed per pass (template keeps the
  * accumulators in VGPRs).  Each block walks tiles grid-stride; gftbls rows
  * [dest_base, dest_base+D) are staged once per block into LDS and re-read
  * as wave-uniform broadcasts once per (dest, src) per tile. */
 template <int D>
 __global__ __launch_bounds__(kThreads) void ec_encode_kernel(
-    uint64_t part_len, int srcs, int dest_base,
+    uint32_t part_len, int srcs, int dest_base,
     const uint8_t *__restrict__ gftbls_dev,  /* 32*srcs*dests_total */
     const uint64_t *__restrict__ src_ptrs,   /* [stripes][srcs]  */
     const uint64_t *__restrict__ dst_ptrs,   /* [stripes][dests_total] */
-    int dests_total, uint32_t tiles_per_part, uint64_t total_tiles) {
+    int dests_total, uint32_t tiles_per_part, uint32_t total_tiles) {
 	extern __shared__ __attribute__((aligned(16))) uint8_t smem[];
-	const int tid = threadIdx.x;
+	const uint32_t tid = threadIdx.x;
 
 	/* Stage this pass's table rows: [D][srcs][32] bytes. */
 	{
@@ -90,10 +106,13 @@ __global__ __launch_bounds__(kThreads) void ec_encode_kernel(
 	}
 	__syncthreads();
 
-	for (uint64_t tile = blockIdx.x; tile < total_tiles; tile += gridDim.x) {
-		uint32_t stripe = (uint32_t)(tile / tiles_per_part);
-		uint32_t t_in_part = (uint32_t)(tile - (uint64_t)stripe * tiles_per_part);
-		uint64_t base = (uint64_t)t_in_part * kTileBytes + (uint32_t)tid * 16u;
+	for (uint32_t tile = blockIdx.x; tile < total_tiles; tile += gridDim.x) {
+		uint32_t stripe = tile / tiles_per_part;
+		uint32_t tin = tile - stripe * tiles_per_part;
+		uint32_t base = tin * kTileBytes + tid * 16u;
+		const uint64_t *sp_tab = src_ptrs + (uint64_t)stripe * srcs;
+		const uint64_t *dp_tab =
+		    dst_ptrs + (uint64_t)stripe * dests_total + dest_base;
 
 		uint4 acc[D][kChunks];
 #pragma unroll
@@ -102,41 +121,71 @@ __global__ __launch_bounds__(kThreads) void ec_encode_kernel(
 			for (int c = 0; c < kChunks; ++c)
 				acc[d][c] = make_uint4(0, 0, 0, 0);
 
-		for (int j = 0; j < srcs; ++j) {
-			const uint8_t *sp =
-			    (const uint8_t *)src_ptrs[(uint64_t)stripe * srcs + j];
-			/* per-(d,j) coefficient tables: wave-uniform LDS broadcast */
-			uint4 L[D], H[D];
+		if ((tin + 1) * kTileBytes <= part_len) {
+			/* full tile: branchless, next source prefetched while the
+			 * current one is accumulated */
+			uint4 w[kChunks], wn[kChunks];
+			{
+				const uint8_t *sp = (const uint8_t *)sp_tab[0];
 #pragma unroll
-			for (int d = 0; d < D; ++d) {
-				const uint8_t *tb = smem + ((size_t)d * srcs + j) * 32;
-				L[d] = *(const uint4 *)tb;
-				H[d] = *(const uint4 *)(tb + 16);
+				for (int c = 0; c < kChunks; ++c)
+					w[c] = *(const uint4 *)(sp + (base + c * kChunkBytes));
+			}
+			for (int j = 0; j < srcs; ++j) {
+				if (j + 1 < srcs) {
+					const uint8_t *spn = (const uint8_t *)sp_tab[j + 1];
+#pragma unroll
+					for (int c = 0; c < kChunks; ++c)
+						wn[c] = *(const uint4 *)(spn + (base + c * kChunkBytes));
+				}
+				uint4 L[D], H[D];
+#pragma unroll
+				for (int d = 0; d < D; ++d) {
+					const uint8_t *tb = smem + ((size_t)d * srcs + j) * 32;
+					L[d] = *(const uint4 *)tb;
+					H[d] = *(const uint4 *)(tb + 16);
+				}
+#pragma unroll
+				for (int c = 0; c < kChunks; ++c)
+					gf_macc_all<D>(acc, c, w[c], L, H);
+#pragma unroll
+				for (int c = 0; c < kChunks; ++c) w[c] = wn[c];
 			}
 #pragma unroll
-			for (int c = 0; c < kChunks; ++c) {
-				uint64_t off = base + (uint32_t)c * (kThreads * 16);
-				if (off < part_len) {
-					uint4 w = *(const uint4 *)(sp + off);
+			for (int d = 0; d < D; ++d) {
+				uint8_t *dp = (uint8_t *)dp_tab[d];
 #pragma unroll
-					for (int d = 0; d < D; ++d) {
-						acc[d][c].x = gf_macc(acc[d][c].x, w.x, L[d], H[d]);
-						acc[d][c].y = gf_macc(acc[d][c].y, w.y, L[d], H[d]);
-						acc[d][c].z = gf_macc(acc[d][c].z, w.z, L[d], H[d]);
-						acc[d][c].w = gf_macc(acc[d][c].w, w.w, L[d], H[d]);
+				for (int c = 0; c < kChunks; ++c)
+					*(uint4 *)(dp + (base + c * kChunkBytes)) = acc[d][c];
+			}
+		} else {
+			/* ragged tail tile: per-strip bounds checks */
+			for (int j = 0; j < srcs; ++j) {
+				const uint8_t *sp = (const uint8_t *)sp_tab[j];
+				uint4 L[D], H[D];
+#pragma unroll
+				for (int d = 0; d < D; ++d) {
+					const uint8_t *tb = smem + ((size_t)d * srcs + j) * 32;
+					L[d] = *(const uint4 *)tb;
+					H[d] = *(const uint4 *)(tb + 16);
+				}
+#pragma unroll
+				for (int c = 0; c < kChunks; ++c) {
+					uint32_t off = base + c * kChunkBytes;
+					if (off < part_len) {
+						uint4 w = *(const uint4 *)(sp + off);
+						gf_macc_all<D>(acc, c, w, L, H);
 					}
 				}
 			}
-		}
-
 #pragma unroll
-		for (int d = 0; d < D; ++d) {
-			uint8_t *dp = (uint8_t *)dst_ptrs[(uint64_t)stripe * dests_total +
-			                                  dest_base + d];
+			for (int d = 0; d < D; ++d) {
+				uint8_t *dp = (uint8_t *)dp_tab[d];
 #pragma unroll
-			for (int c = 0; c < kChunks; ++c) {
-				uint64_t off = base + (uint32_t)c * (kThreads * 16);
-				if (off < part_len) *(uint4 *)(dp + off) = acc[d][c];
+				for (int c = 0; c < kChunks; ++c) {
+					uint32_t off = base + c * kChunkBytes;
+					if (off < part_len) *(uint4 *)(dp + off) = acc[d][c];
+				}
 			}
 		}
 	}
@@ -231,16 +280,31 @@ __global__ __launch_bounds__(kThreads) void crc32_blocks_kernel(
 }
 
 /* ------------------------------------------------------------------ */
-/* Engine                                                             */
+/* Engine + plans                                                     */
 /* ------------------------------------------------------------------ */
 
 struct lizec_engine {
 	int device;
 	hipStream_t stream;        /* default stream for calls passing NULL */
-	uint8_t *d_gftbls;         /* 32*32*32 max */
-	uint64_t *d_ptrs;          /* grows */
+	uint8_t *d_gftbls;         /* 32*32*32 max (per-call scratch) */
+	uint64_t *d_ptrs;          /* per-call scratch, grows */
 	size_t ptrs_cap;           /* in elements */
 	uint32_t *d_crc_const;     /* kCrcConstWords */
+};
+
+/* A plan = a prepared batch: device-resident tables + pointer arrays.
+ * Mirrors the reference's cached-matrix + read-plan structure
+ * (reed_solomon.h:194-198, read_plan.h): build once per (erasure pattern,
+ * batch), run many times with zero host->device traffic. */
+struct lizec_plan {
+	lizec_engine *e;
+	uint64_t part_len;
+	int srcs, dests, num_stripes;
+	uint32_t tiles_per_part;
+	uint32_t total_tiles;
+	uint8_t *d_tbls;
+	uint64_t *d_src;
+	uint64_t *d_dst;
 };
 
 extern "C" int lizec_gpu_count(void) {
@@ -317,10 +381,10 @@ extern "C" int lizec_engine_create(lizec_engine **out, int device_id) {
 
 extern "C" void lizec_engine_destroy(lizec_engine *e) {
 	if (!e) return;
-	hipFree(e->d_gftbls);
-	hipFree(e->d_ptrs);
-	hipFree(e->d_crc_const);
-	hipStreamDestroy(e->stream);
+	(void)hipFree(e->d_gftbls);
+	(void)hipFree(e->d_ptrs);
+	(void)hipFree(e->d_crc_const);
+	(void)hipStreamDestroy(e->stream);
 	delete e;
 }
 
@@ -342,17 +406,53 @@ static int ensure_ptrs(lizec_engine *e, size_t n) {
 	return LIZEC_OK;
 }
 
+static int check_batch_args(uint64_t part_len, int srcs, int dests,
+                            int num_stripes) {
+	if (srcs < 1 || srcs > 32 || dests < 1 || dests > 32 || num_stripes < 1)
+		return LIZEC_EINVAL;
+	if (part_len == 0 || (part_len & 15) || part_len > (uint64_t)1 << 31)
+		return LIZEC_EINVAL;
+	return LIZEC_OK;
+}
+
 template <int D>
-static void launch_ec(uint64_t part_len, int srcs, int dest_base,
+static void launch_ec(uint32_t part_len, int srcs, int dest_base,
                       const uint8_t *d_tbls, const uint64_t *d_src,
                       const uint64_t *d_dst, int dests_total,
-                      uint32_t tiles_per_part, uint64_t total_tiles,
+                      uint32_t tiles_per_part, uint32_t total_tiles,
                       hipStream_t s) {
-	uint32_t grid = (uint32_t)(total_tiles < 262144 ? total_tiles : 262144);
+	uint32_t grid = total_tiles < 262144u ? total_tiles : 262144u;
 	size_t lds = (size_t)D * srcs * 32;
 	hipLaunchKernelGGL(ec_encode_kernel<D>, dim3(grid), dim3(kThreads), lds, s,
 	                   part_len, srcs, dest_base, d_tbls, d_src, d_dst,
 	                   dests_total, tiles_per_part, total_tiles);
+}
+
+static int run_batch(uint64_t part_len, int srcs, int dests,
+                     const uint8_t *d_tbls, const uint64_t *d_src,
+                     const uint64_t *d_dst, uint32_t tiles_per_part,
+                     uint32_t total_tiles, hipStream_t s) {
+	for (int base = 0; base < dests;) {
+		int d = dests - base;
+		if (d > 4) d = 4;
+		switch (d) {
+		case 1: launch_ec<1>((uint32_t)part_len, srcs, base, d_tbls, d_src,
+		                     d_dst, dests, tiles_per_part, total_tiles, s);
+			break;
+		case 2: launch_ec<2>((uint32_t)part_len, srcs, base, d_tbls, d_src,
+		                     d_dst, dests, tiles_per_part, total_tiles, s);
+			break;
+		case 3: launch_ec<3>((uint32_t)part_len, srcs, base, d_tbls, d_src,
+		                     d_dst, dests, tiles_per_part, total_tiles, s);
+			break;
+		default: launch_ec<4>((uint32_t)part_len, srcs, base, d_tbls, d_src,
+		                      d_dst, dests, tiles_per_part, total_tiles, s);
+			break;
+		}
+		base += d;
+	}
+	LIZEC_CHECK(hipGetLastError());
+	return LIZEC_OK;
 }
 
 extern "C" int lizec_ec_encode_batch(lizec_engine *e, uint64_t part_len,
@@ -362,15 +462,14 @@ extern "C" int lizec_ec_encode_batch(lizec_engine *e, uint64_t part_len,
                                      const uint64_t *dst_dptrs,
                                      int num_stripes, void *stream) {
 	if (!e) return LIZEC_EINVAL;
-	if (srcs < 1 || srcs > 32 || dests < 1 || dests > 32 || num_stripes < 1)
-		return LIZEC_EINVAL;
-	if (part_len == 0 || (part_len & 15)) return LIZEC_EINVAL;
+	int r = check_batch_args(part_len, srcs, dests, num_stripes);
+	if (r != LIZEC_OK) return r;
 	hipStream_t s = stream ? (hipStream_t)stream : e->stream;
 	LIZEC_CHECK(hipSetDevice(e->device));
 
 	size_t nsrc = (size_t)num_stripes * srcs;
 	size_t ndst = (size_t)num_stripes * dests;
-	int r = ensure_ptrs(e, nsrc + ndst);
+	r = ensure_ptrs(e, nsrc + ndst);
 	if (r != LIZEC_OK) return r;
 	uint64_t *d_src = e->d_ptrs;
 	uint64_t *d_dst = e->d_ptrs + nsrc;
@@ -381,26 +480,67 @@ extern "C" int lizec_ec_encode_batch(lizec_engine *e, uint64_t part_len,
 	LIZEC_CHECK(hipMemcpyAsync(d_dst, dst_dptrs, ndst * 8,
 	                           hipMemcpyHostToDevice, s));
 
-	uint32_t tiles_per_part = (uint32_t)((part_len + kTileBytes - 1) / kTileBytes);
-	uint64_t total_tiles = (uint64_t)tiles_per_part * num_stripes;
+	uint32_t tiles_per_part =
+	    (uint32_t)((part_len + kTileBytes - 1) / kTileBytes);
+	uint32_t total_tiles = tiles_per_part * (uint32_t)num_stripes;
+	return run_batch(part_len, srcs, dests, e->d_gftbls, d_src, d_dst,
+	                 tiles_per_part, total_tiles, s);
+}
 
-	for (int base = 0; base < dests;) {
-		int d = dests - base;
-		if (d > 4) d = 4;
-		switch (d) {
-		case 1: launch_ec<1>(part_len, srcs, base, e->d_gftbls, d_src, d_dst,
-		                     dests, tiles_per_part, total_tiles, s); break;
-		case 2: launch_ec<2>(part_len, srcs, base, e->d_gftbls, d_src, d_dst,
-		                     dests, tiles_per_part, total_tiles, s); break;
-		case 3: launch_ec<3>(part_len, srcs, base, e->d_gftbls, d_src, d_dst,
-		                     dests, tiles_per_part, total_tiles, s); break;
-		default: launch_ec<4>(part_len, srcs, base, e->d_gftbls, d_src, d_dst,
-		                      dests, tiles_per_part, total_tiles, s); break;
-		}
-		base += d;
+extern "C" int lizec_ec_plan_create(lizec_engine *e, uint64_t part_len,
+                                    int srcs, int dests,
+                                    const uint8_t *gftbls,
+                                    const uint64_t *src_dptrs,
+                                    const uint64_t *dst_dptrs,
+                                    int num_stripes, lizec_plan **out) {
+	*out = nullptr;
+	if (!e) return LIZEC_EINVAL;
+	int r = check_batch_args(part_len, srcs, dests, num_stripes);
+	if (r != LIZEC_OK) return r;
+	LIZEC_CHECK(hipSetDevice(e->device));
+	lizec_plan *p = new lizec_plan();
+	p->e = e;
+	p->part_len = part_len;
+	p->srcs = srcs;
+	p->dests = dests;
+	p->num_stripes = num_stripes;
+	p->tiles_per_part = (uint32_t)((part_len + kTileBytes - 1) / kTileBytes);
+	p->total_tiles = p->tiles_per_part * (uint32_t)num_stripes;
+	size_t nsrc = (size_t)num_stripes * srcs;
+	size_t ndst = (size_t)num_stripes * dests;
+	if (hipMalloc(&p->d_tbls, (size_t)32 * srcs * dests) != hipSuccess ||
+	    hipMalloc(&p->d_src, nsrc * 8) != hipSuccess ||
+	    hipMalloc(&p->d_dst, ndst * 8) != hipSuccess) {
+		lizec_ec_plan_destroy(p);
+		return LIZEC_ENOMEM;
 	}
-	LIZEC_CHECK(hipGetLastError());
+	if (hipMemcpy(p->d_tbls, gftbls, (size_t)32 * srcs * dests,
+	              hipMemcpyHostToDevice) != hipSuccess ||
+	    hipMemcpy(p->d_src, src_dptrs, nsrc * 8,
+	              hipMemcpyHostToDevice) != hipSuccess ||
+	    hipMemcpy(p->d_dst, dst_dptrs, ndst * 8,
+	              hipMemcpyHostToDevice) != hipSuccess) {
+		lizec_ec_plan_destroy(p);
+		return LIZEC_EHIP;
+	}
+	*out = p;
 	return LIZEC_OK;
+}
+
+extern "C" int lizec_ec_plan_run(lizec_plan *p, void *stream) {
+	if (!p) return LIZEC_EINVAL;
+	hipStream_t s = stream ? (hipStream_t)stream : p->e->stream;
+	LIZEC_CHECK(hipSetDevice(p->e->device));
+	return run_batch(p->part_len, p->srcs, p->dests, p->d_tbls, p->d_src,
+	                 p->d_dst, p->tiles_per_part, p->total_tiles, s);
+}
+
+extern "C" void lizec_ec_plan_destroy(lizec_plan *p) {
+	if (!p) return;
+	(void)hipFree(p->d_tbls);
+	(void)hipFree(p->d_src);
+	(void)hipFree(p->d_dst);
+	delete p;
 }
 
 extern "C" int lizec_crc32_batch(lizec_engine *e, const void *dev_buf,

@@ -35,8 +35,16 @@ class ReedSolomon:
         self.device = device
         self.slice_type = slice_traits.ec_slice_type(k, m)
         self._tables = {}
+        self._plans = {}
         self._engine = L.engine(device)
         self._lib = L.lib()
+
+    def __del__(self):
+        try:
+            for p in getattr(self, "_plans", {}).values():
+                self._lib.lizec_ec_plan_destroy(p)
+        except Exception:
+            pass
 
     # ---------------- tables (host-side matrix algebra) ---
 
@@ -58,8 +66,27 @@ class ReedSolomon:
 
     # ---------------- batch ops ----------------
 
-    def _run(self, part_len, tbl, ic, oc, src_ptrs, dst_ptrs, nstripes):
+    def _run(self, part_len, tbl, ic, oc, src_ptrs, dst_ptrs, nstripes,
+             plan_key=None):
+        """Launch a batch.  When plan_key is given (stable buffers), the
+        device-side tables+pointer arrays are built once and reused
+        (lizec_ec_plan_*), so steady-state steps do zero host->device
+        traffic — mirroring the reference's matrix cache."""
         stream = torch.cuda.current_stream(self.device).cuda_stream
+        if plan_key is not None:
+            plan = self._plans.get(plan_key)
+            if plan is None:
+                plan = ctypes.c_void_p()
+                L.check(self._lib.lizec_ec_plan_create(
+                    self._engine, part_len, ic, oc,
+                    tbl.ctypes.data_as(ctypes.POINTER(ctypes.c_uint8)),
+                    src_ptrs.ctypes.data_as(ctypes.POINTER(ctypes.c_uint64)),
+                    dst_ptrs.ctypes.data_as(ctypes.POINTER(ctypes.c_uint64)),
+                    nstripes, ctypes.byref(plan)), "lizec_ec_plan_create")
+                self._plans[plan_key] = plan
+            L.check(self._lib.lizec_ec_plan_run(plan, ctypes.c_void_p(stream)),
+                    "lizec_ec_plan_run")
+            return
         L.check(self._lib.lizec_ec_encode_batch(
             self._engine, part_len, ic, oc,
             tbl.ctypes.data_as(ctypes.POINTER(ctypes.c_uint8)),
@@ -101,7 +128,8 @@ class ReedSolomon:
                np.arange(S * k, dtype=np.uint64) * np.uint64(plen))
         dst = (parity.data_ptr() +
                np.arange(S * self.m, dtype=np.uint64) * np.uint64(plen))
-        self._run(plen, tbl, ic, oc, src, dst, S)
+        plan_key = ("enc", data.data_ptr(), parity.data_ptr(), S, plen)
+        self._run(plen, tbl, ic, oc, src, dst, S, plan_key=plan_key)
         return parity
 
     def recover_batch(self, fragments, erased, want=None):

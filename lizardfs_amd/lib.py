@@ -73,6 +73,14 @@ def lib():
         ctypes.c_void_p, ctypes.c_uint64, ctypes.c_int, ctypes.c_int, u8p,
         ctypes.POINTER(ctypes.c_uint64), ctypes.POINTER(ctypes.c_uint64),
         ctypes.c_int, ctypes.c_void_p]
+    L.lizec_ec_plan_create.restype = ctypes.c_int
+    L.lizec_ec_plan_create.argtypes = [
+        ctypes.c_void_p, ctypes.c_uint64, ctypes.c_int, ctypes.c_int, u8p,
+        ctypes.POINTER(ctypes.c_uint64), ctypes.POINTER(ctypes.c_uint64),
+        ctypes.c_int, ctypes.POINTER(ctypes.c_void_p)]
+    L.lizec_ec_plan_run.restype = ctypes.c_int
+    L.lizec_ec_plan_run.argtypes = [ctypes.c_void_p, ctypes.c_void_p]
+    L.lizec_ec_plan_destroy.argtypes = [ctypes.c_void_p]
     L.lizec_crc32_batch.restype = ctypes.c_int
     L.lizec_crc32_batch.argtypes = [
         ctypes.c_void_p, ctypes.c_void_p, ctypes.c_uint32, ctypes.c_uint64,
