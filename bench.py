@@ -8,16 +8,20 @@ the whole batch.  value = whole-job data-in GiB/s across all ranks
 (the reference's own throughput convention, reed_solomon_unittest.cc:44-73).
 
 Usage: python bench.py [--gpus N] [--steps K] [--warmup W] [--stripes S]
+                       [--op encode|decode|crc]
 N>1 is launched by the driver via torch.distributed.run (one rank per GPU,
 RCCL); stripes shard across ranks as independent batches (weak scaling — the
-reference processes stripes independently too; no data-path collective).
+reference processes stripes independently too; no data-path collective,
+SURVEY §8e).
+
+--op decode measures BASELINE config 3 (2 erased data parts); --op crc the
+per-64KiB-block CRC32 gate.  The default (encode) is the contract line.
 
 The cpu_baseline leg times the ORACLE (CPU restatement of the reference
 path, OpenMP over stripes) on a bounded sample — reported context, not the
 roofline target.
 """
 import argparse
-import ctypes
 import json
 import os
 import sys
@@ -34,42 +38,68 @@ K_EC = 8
 M_EC = 2
 STRIPE_BYTES = 64 * 1024 * 1024
 PART_LEN = STRIPE_BYTES // K_EC
+ERASED = (1, 5)   # BASELINE config 3: 2 erased data parts
 
 
 def log(msg):
     print(msg, file=sys.stderr, flush=True)
 
 
-def cpu_baseline_leg(target_seconds=12.0):
-    """Time the oracle's threaded encode on a bounded sample of the same
+def cpu_baseline_leg(op, target_seconds=10.0):
+    """Time the oracle's threaded path on a bounded sample of the same
     workload on this box's host cores.  kind='port' (restatement of
-    galois_field_encode.cc:28-47 + reed_solomon.h encode semantics)."""
+    galois_field_encode.cc:28-47 + reed_solomon.h semantics)."""
     import oracle
     cores = int(os.environ.get("OMP_NUM_THREADS", os.cpu_count() or 1))
-    tbl, ic, oc = oracle.rs_make_tables(
-        K_EC, M_EC, (1 << K_EC) - 1, (1 << K_EC) - 1,
-        ((1 << M_EC) - 1) << K_EC)
-    # calibrate with one stripe, then size the sample
     rng = np.random.default_rng(42)
-    data = rng.integers(0, 256, (1, K_EC, PART_LEN), np.uint8)
-    parity = np.zeros((1, M_EC, PART_LEN), np.uint8)
-    t0 = time.perf_counter()
-    oracle.encode_stripes(K_EC, M_EC, PART_LEN, 1, tbl, data, parity)
-    per_stripe = time.perf_counter() - t0
-    n = max(2, min(64, int(target_seconds / max(per_stripe, 1e-3))))
+    n = 16  # 1 GiB data sample, repeated until ~target_seconds
     data = rng.integers(0, 256, (n, K_EC, PART_LEN), np.uint8)
-    parity = np.zeros((n, M_EC, PART_LEN), np.uint8)
+
+    if op == "crc":
+        buf = np.ascontiguousarray(data.reshape(-1))
+        t0 = time.perf_counter()
+        oracle.crc32_blocks(buf, 65536)
+        once = time.perf_counter() - t0
+        reps = max(1, int(target_seconds / max(once, 1e-3)))
+        t0 = time.perf_counter()
+        for _ in range(reps):
+            oracle.crc32_blocks(buf, 65536)
+        dt = time.perf_counter() - t0
+        gib = reps * buf.size / (1 << 30)
+        sample = f"{reps}x{n * 64}MiB CRC32/64KiB blocks, {dt:.1f}s, OpenMP"
+        return {"value": round(gib / dt, 3), "unit": "GiB/s", "cores": cores,
+                "kind": "port", "sample": sample}
+
+    if op == "decode":
+        present = sum(1 << i for i in range(K_EC + M_EC) if i not in ERASED)
+        needed = sum(1 << i for i in ERASED)
+        tbl, ic, oc = oracle.rs_make_tables(K_EC, M_EC, present, present,
+                                            needed)
+        # reuse encode_stripes shape: inputs = the ic surviving parts
+        parity = np.zeros((n, oc, PART_LEN), np.uint8)
+        srcs = np.ascontiguousarray(data[:, :ic])  # ic=8 surviving parts
+        enc = lambda: oracle.encode_stripes(ic, oc, PART_LEN, n, tbl,
+                                            srcs, parity)
+    else:
+        tbl, ic, oc = oracle.rs_make_tables(
+            K_EC, M_EC, (1 << K_EC) - 1, (1 << K_EC) - 1,
+            ((1 << M_EC) - 1) << K_EC)
+        parity = np.zeros((n, M_EC, PART_LEN), np.uint8)
+        enc = lambda: oracle.encode_stripes(K_EC, M_EC, PART_LEN, n, tbl,
+                                            data, parity)
+
     t0 = time.perf_counter()
-    oracle.encode_stripes(K_EC, M_EC, PART_LEN, n, tbl, data, parity)
+    enc()
+    once = time.perf_counter() - t0
+    reps = max(1, int(target_seconds / max(once, 1e-3)))
+    t0 = time.perf_counter()
+    for _ in range(reps):
+        enc()
     dt = time.perf_counter() - t0
-    gib = n * STRIPE_BYTES / (1 << 30)
-    return {
-        "value": round(gib / dt, 3),
-        "unit": "GiB/s",
-        "cores": cores,
-        "kind": "port",
-        "sample": f"{n}x64MiB stripes ec(8,2) encode, {dt:.1f}s, OpenMP",
-    }
+    gib = reps * n * STRIPE_BYTES / (1 << 30)
+    sample = f"{reps}x{n}x64MiB stripes ec(8,2) {op}, {dt:.1f}s, OpenMP"
+    return {"value": round(gib / dt, 3), "unit": "GiB/s", "cores": cores,
+            "kind": "port", "sample": sample}
 
 
 def read_traffic_calibration(workload):
@@ -78,9 +108,10 @@ def read_traffic_calibration(workload):
     path = os.path.join(REPO, "profiles", "pmc_traffic.json")
     try:
         with open(path) as f:
-            d = json.load(f)
-        if d.get("workload") == workload:
-            return d.get("bytes_per_launch")
+            entries = json.load(f)
+        for d in entries if isinstance(entries, list) else [entries]:
+            if d.get("workload") == workload:
+                return d.get("bytes_per_launch")
     except Exception:
         pass
     return None
@@ -93,6 +124,8 @@ def main():
     ap.add_argument("--warmup", type=int, default=5)
     ap.add_argument("--stripes", type=int, default=1024,
                     help="stripes per rank (weak scaling)")
+    ap.add_argument("--op", choices=("encode", "decode", "crc"),
+                    default="encode")
     ap.add_argument("--skip-cpu-baseline", action="store_true")
     args = ap.parse_args()
 
@@ -111,23 +144,51 @@ def main():
         torch.distributed.init_process_group("nccl")
 
     from lizardfs_amd.ec import ReedSolomon
+    from lizardfs_amd import crc as lcrc
 
     S = args.stripes
-    workload = (f"ec({K_EC},{M_EC}) encode, {S}x64MiB synthetic stripes/GPU, "
-                f"device-resident")
+    workload = (f"ec({K_EC},{M_EC}) {args.op}, {S}x64MiB synthetic "
+                f"stripes/GPU, device-resident")
     log(f"[rank {rank}] generating {S} stripes "
         f"({S * STRIPE_BYTES / (1 << 30):.0f} GiB data) on cuda:{local_rank}")
     g = torch.Generator(device="cuda").manual_seed(42 + rank)
     data = torch.randint(0, 256, (S, K_EC, PART_LEN), dtype=torch.uint8,
                          device="cuda", generator=g)
-    parity = torch.empty((S, M_EC, PART_LEN), dtype=torch.uint8, device="cuda")
+    parity = torch.empty((S, M_EC, PART_LEN), dtype=torch.uint8,
+                         device="cuda")
 
     rs = ReedSolomon(K_EC, M_EC, device=local_rank)
+    nparts = K_EC + M_EC
 
-    def step():
-        rs.encode_batch(data, parity)
+    if args.op == "encode":
+        def step():
+            rs.encode_batch(data, parity)
+        # dominant kernel: ec_encode_kernel<2>; 1 launch/step
+        alg_bytes_per_launch = S * STRIPE_BYTES * (K_EC + M_EC) // K_EC
+        metric = "GiB/s EC encode, ec(8,2) 64MiB stripes"
+    elif args.op == "decode":
+        rs.encode_batch(data, parity)   # produce real parity first
+        rs.sync()
+        frags = [None if i in ERASED else
+                 (data[:, i, :] if i < K_EC else parity[:, i - K_EC, :])
+                 for i in range(nparts)]
+        outs = {i: torch.empty((S, PART_LEN), dtype=torch.uint8,
+                               device="cuda") for i in ERASED}
 
-    # warmup
+        def step():
+            rs.recover_batch(frags, erased=ERASED, out=outs)
+        alg_bytes_per_launch = S * PART_LEN * (K_EC + len(ERASED))
+        metric = "GiB/s EC decode(2 erasures), ec(8,2) 64MiB stripes"
+    else:  # crc
+        flat = data.reshape(-1)
+        crcs = torch.empty(flat.numel() // 65536, dtype=torch.int32,
+                           device="cuda")
+
+        def step():
+            lcrc.crc32_blocks(flat, 65536, out=crcs)
+        alg_bytes_per_launch = S * STRIPE_BYTES + 4 * (S * STRIPE_BYTES // 65536)
+        metric = "GiB/s CRC32 per-64KiB-block"
+
     for _ in range(args.warmup):
         step()
     torch.cuda.synchronize()
@@ -163,10 +224,6 @@ def main():
     if rank != 0:
         return
 
-    # roofline for the dominant kernel (ec_encode_kernel<2>): one launch per
-    # step; algorithmic bytes = read k*L + write m*L per stripe (SURVEY §8d:
-    # ec(8,2) moves 1.25 B per data byte).
-    alg_bytes_per_launch = S * STRIPE_BYTES * (K_EC + M_EC) // K_EC
     achieved_gbps = alg_bytes_per_launch / (avg_launch_ms / 1e3) / 1e9
     peak_gbps = 8000.0  # MI355X_MICROARCH.md: HBM3E 8 TB/s spec peak
     traffic = read_traffic_calibration(workload)
@@ -174,10 +231,10 @@ def main():
     cpu_baseline = None
     if n_gpus == 1 and not args.skip_cpu_baseline:
         log("[rank 0] timing CPU baseline (oracle, OpenMP)...")
-        cpu_baseline = cpu_baseline_leg()
+        cpu_baseline = cpu_baseline_leg(args.op)
 
     result = {
-        "metric": "GiB/s EC encode, ec(8,2) 64MiB stripes",
+        "metric": metric,
         "value": round(value, 2),
         "unit": "GiB/s",
         "n_gpus": n_gpus,

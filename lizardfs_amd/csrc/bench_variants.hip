@@ -1,0 +1,164 @@
+/* bench_variants.hip — A/B harness for the EC encode kernel variants on a
+ * real MI355X.  Standalone executable (not part of liblizec.so): allocates
+ * device-resident ec(8,2)-shaped batches, runs each template configuration,
+ * checks a sample of the output against the host scalar path, and prints a
+ * GB/s table (traffic = (k+m)/k bytes per data byte).
+ *
+ * Build: make bench_variants  ; run on the GPU box only.
+ */
+#include <hip/hip_runtime.h>
+
+#include <cstdint>
+#include <cstdio>
+#include <cstdlib>
+#include <cstring>
+#include <vector>
+
+#include "../../include/lizec.h"
+#include "ec_kernel.h"
+
+#define CK(x)                                                           \
+	do {                                                                \
+		hipError_t e = (x);                                             \
+		if (e != hipSuccess) {                                          \
+			fprintf(stderr, "HIP error %s at %s:%d\n",                  \
+			        hipGetErrorString(e), __FILE__, __LINE__);          \
+			exit(1);                                                    \
+		}                                                               \
+	} while (0)
+
+__global__ void fill_kernel(uint8_t *p, size_t n, uint32_t salt) {
+	size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+	size_t stride = (size_t)gridDim.x * blockDim.x;
+	for (; i * 4 < n; i += stride) {
+		uint32_t v = (uint32_t)(i * 2654435761u) ^ salt;
+		v ^= v >> 15;
+		v *= 0x2c1b3c6du;
+		v ^= v >> 12;
+		((uint32_t *)p)[i] = v;
+	}
+}
+
+struct Variant {
+	const char *name;
+	void (*launch)(uint32_t, int, const uint8_t *, const uint64_t *,
+	               const uint64_t *, int, uint32_t, uint32_t, uint32_t,
+	               hipStream_t);
+	int ch;
+};
+
+template <int D, int CH, bool SWZ, bool NTST>
+static void launch_var(uint32_t part_len, int srcs, const uint8_t *tbls,
+                       const uint64_t *src, const uint64_t *dst, int dests,
+                       uint32_t tiles_per_part, uint32_t total_tiles,
+                       uint32_t grid_cap, hipStream_t s) {
+	uint32_t grid = total_tiles < grid_cap ? total_tiles : grid_cap;
+	size_t lds = (size_t)D * srcs * 32;
+	hipLaunchKernelGGL(HIP_KERNEL_NAME(ec_encode_kernel<D, CH, SWZ, NTST>),
+	                   dim3(grid), dim3(kThreads), lds, s, part_len, srcs, 0,
+	                   tbls, src, dst, dests, tiles_per_part, total_tiles);
+}
+
+int main(int argc, char **argv) {
+	int k = 8, m = 2, stripes = 1024, reps = 6;
+	uint64_t part_len = 8u * 1024 * 1024;
+	if (argc > 1) stripes = atoi(argv[1]);
+	if (argc > 2) reps = atoi(argv[2]);
+
+	size_t data_bytes = (size_t)stripes * k * part_len;
+	size_t par_bytes = (size_t)stripes * m * part_len;
+	printf("ec(%d,%d) stripes=%d part=%lu MiB data=%.1f GiB\n", k, m, stripes,
+	       part_len >> 20, data_bytes / 1073741824.0);
+
+	uint8_t *d_data, *d_par;
+	CK(hipMalloc(&d_data, data_bytes));
+	CK(hipMalloc(&d_par, par_bytes));
+	hipLaunchKernelGGL(fill_kernel, dim3(4096), dim3(256), 0, 0, d_data,
+	                   data_bytes, 0x1234567u);
+
+	uint8_t tbls[32 * 32 * 2];
+	if (lizec_rs_encode_tables(k, m, tbls) != 0) return 1;
+	uint8_t *d_tbls;
+	CK(hipMalloc(&d_tbls, 32 * k * m));
+	CK(hipMemcpy(d_tbls, tbls, 32 * k * m, hipMemcpyHostToDevice));
+
+	std::vector<uint64_t> sp(stripes * k), dp(stripes * m);
+	for (int s = 0; s < stripes; ++s) {
+		for (int j = 0; j < k; ++j)
+			sp[s * k + j] = (uint64_t)(d_data + ((size_t)s * k + j) * part_len);
+		for (int l = 0; l < m; ++l)
+			dp[s * m + l] = (uint64_t)(d_par + ((size_t)s * m + l) * part_len);
+	}
+	uint64_t *d_sp, *d_dp;
+	CK(hipMalloc(&d_sp, sp.size() * 8));
+	CK(hipMalloc(&d_dp, dp.size() * 8));
+	CK(hipMemcpy(d_sp, sp.data(), sp.size() * 8, hipMemcpyHostToDevice));
+	CK(hipMemcpy(d_dp, dp.data(), dp.size() * 8, hipMemcpyHostToDevice));
+
+	/* host reference for a sample window */
+	const int SAMPLE = 4096;
+	std::vector<uint8_t> h_src(k * SAMPLE), h_exp(m * SAMPLE), h_got(SAMPLE);
+	{
+		for (int j = 0; j < k; ++j)
+			CK(hipMemcpy(&h_src[j * SAMPLE], (void *)sp[j], SAMPLE,
+			             hipMemcpyDeviceToHost));
+		uint8_t *srcp[32], *dstp[32];
+		for (int j = 0; j < k; ++j) srcp[j] = &h_src[j * SAMPLE];
+		for (int l = 0; l < m; ++l) dstp[l] = &h_exp[l * SAMPLE];
+		ec_encode_data(SAMPLE, k, m, tbls, srcp, dstp);
+	}
+
+	struct Cfg {
+		const char *name;
+		void (*fn)(uint32_t, int, const uint8_t *, const uint64_t *,
+		           const uint64_t *, int, uint32_t, uint32_t, uint32_t,
+		           hipStream_t);
+		int ch;
+		uint32_t grid_cap;
+	};
+	Cfg cfgs[] = {
+	    {"D2_CH4_base      ", launch_var<2, 4, false, false>, 4, 262144},
+	    {"D2_CH4_swz       ", launch_var<2, 4, true, false>, 4, 262144},
+	    {"D2_CH4_nt        ", launch_var<2, 4, false, true>, 4, 262144},
+	    {"D2_CH4_swz_nt    ", launch_var<2, 4, true, true>, 4, 262144},
+	    {"D2_CH8_base      ", launch_var<2, 8, false, false>, 8, 262144},
+	    {"D2_CH8_swz_nt    ", launch_var<2, 8, true, true>, 8, 262144},
+	    {"D2_CH2_base      ", launch_var<2, 2, false, false>, 2, 262144},
+	    {"D2_CH4_grid16k   ", launch_var<2, 4, false, false>, 4, 16384},
+	    {"D2_CH4_swz_g16k  ", launch_var<2, 4, true, false>, 4, 16384},
+	    {"D2_CH4_nt_g2048  ", launch_var<2, 4, false, true>, 4, 2048},
+	};
+
+	double traffic = (double)stripes * part_len * (k + m);
+	hipEvent_t e0, e1;
+	CK(hipEventCreate(&e0));
+	CK(hipEventCreate(&e1));
+
+	for (auto &c : cfgs) {
+		uint32_t tpp = (uint32_t)((part_len + c.ch * kChunkBytes - 1) /
+		                          (c.ch * kChunkBytes));
+		uint32_t tot = tpp * stripes;
+		CK(hipMemset(d_par, 0, 4096));
+		/* warmup x2 */
+		for (int r = 0; r < 2; ++r)
+			c.fn((uint32_t)part_len, k, d_tbls, d_sp, d_dp, m, tpp, tot,
+			     c.grid_cap, 0);
+		CK(hipDeviceSynchronize());
+		CK(hipEventRecord(e0, 0));
+		for (int r = 0; r < reps; ++r)
+			c.fn((uint32_t)part_len, k, d_tbls, d_sp, d_dp, m, tpp, tot,
+			     c.grid_cap, 0);
+		CK(hipEventRecord(e1, 0));
+		CK(hipDeviceSynchronize());
+		float ms;
+		CK(hipEventElapsedTime(&ms, e0, e1));
+		double gbps = traffic * reps / (ms / 1e3) / 1e9;
+		/* verify sample of parity 0 */
+		CK(hipMemcpy(h_got.data(), (void *)dp[0], SAMPLE,
+		             hipMemcpyDeviceToHost));
+		bool ok = memcmp(h_got.data(), h_exp.data(), SAMPLE) == 0;
+		printf("%s  %8.1f GB/s  (%.3f ms/rep)  %s\n", c.name, gbps,
+		       ms / reps, ok ? "OK" : "WRONG");
+	}
+	return 0;
+}

@@ -1,0 +1,182 @@
+/* ec_kernel.h — the GF(2^8) EC encode/decode kernel for gfx950 (CDNA4),
+ * shared between the product library (lizec_gpu.hip) and the variant
+ * benchmark harness (bench_variants.hip).
+ *
+ * Algorithm: dest[l][i] = XOR_j tbl(l,j)[src[j][i]] over GF(2^8) — the
+ * reference's ec_encode_data contract (galois_field_encode.cc:28-47,
+ * ISA-L table layout [dest][src][{lo16,hi16}]).  GF multiply by a constant
+ * is two in-register 16-byte LUTs (low/high nibble products) built from
+ * v_perm_b32; tables are staged in LDS per block and broadcast-read into
+ * VGPRs once per (dest, src) per tile.
+ *
+ * Template knobs:
+ *   D    destinations per pass (accumulators in VGPRs)
+ *   CH   4-KiB chunks per tile (per-thread bytes = CH*16)
+ *   SWZ  XCD-aware block remap (dispatcher puts block b on XCD b%8; the
+ *        bijective remap gives each XCD a contiguous tile range)
+ *   NTST non-temporal parity stores (streaming, no reuse)
+ */
+#pragma once
+#include <hip/hip_runtime.h>
+#include <cstdint>
+
+constexpr int kThreads = 256;
+constexpr uint32_t kChunkBytes = kThreads * 16;   /* 4096 */
+
+/* 16-entry byte LUT on 4 packed nibbles via v_perm_b32 (sel byte 0-3 picks
+ * a byte of src1, 4-7 of src0). */
+__device__ __forceinline__ uint32_t lut16(uint32_t t0, uint32_t t1,
+                                          uint32_t t2, uint32_t t3,
+                                          uint32_t nib) {
+	uint32_t s3 = nib & 0x07070707u;
+	uint32_t lo = __builtin_amdgcn_perm(t1, t0, s3);   /* nib in 0..7  */
+	uint32_t hi = __builtin_amdgcn_perm(t3, t2, s3);   /* nib in 8..15 */
+	uint32_t msel = 0x03020100u | ((nib >> 1) & 0x04040404u);
+	return __builtin_amdgcn_perm(hi, lo, msel);
+}
+
+/* GF(2^8) multiply-accumulate of one 32-bit word against one coefficient
+ * table (L = products of low nibbles, H = of high nibbles). */
+__device__ __forceinline__ uint32_t gf_macc(uint32_t acc, uint32_t w,
+                                            const uint4 &L, const uint4 &H) {
+	uint32_t nl = w & 0x0f0f0f0fu;
+	uint32_t nh = (w >> 4) & 0x0f0f0f0fu;
+	return acc ^ lut16(L.x, L.y, L.z, L.w, nl) ^ lut16(H.x, H.y, H.z, H.w, nh);
+}
+
+template <int D, int CH>
+__device__ __forceinline__ void gf_macc_all(uint4 (&acc)[D][CH], int c,
+                                            const uint4 &w, const uint4 (&L)[D],
+                                            const uint4 (&H)[D]) {
+#pragma unroll
+	for (int d = 0; d < D; ++d) {
+		acc[d][c].x = gf_macc(acc[d][c].x, w.x, L[d], H[d]);
+		acc[d][c].y = gf_macc(acc[d][c].y, w.y, L[d], H[d]);
+		acc[d][c].z = gf_macc(acc[d][c].z, w.z, L[d], H[d]);
+		acc[d][c].w = gf_macc(acc[d][c].w, w.w, L[d], H[d]);
+	}
+}
+
+/* Bijective XCD remap (8 XCDs): consecutive hardware block ids round-robin
+ * the XCDs; remapped ids give each XCD one contiguous range. */
+__device__ __forceinline__ uint32_t xcd_remap(uint32_t b, uint32_t n) {
+	uint32_t xcd = b & 7u, i = b >> 3;
+	uint32_t q = n >> 3, r = n & 7u;
+	return (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + i;
+}
+
+template <int D, int CH, bool SWZ, bool NTST>
+__global__ __launch_bounds__(kThreads) void ec_encode_kernel(
+    uint32_t part_len, int srcs, int dest_base,
+    const uint8_t *__restrict__ gftbls_dev,  /* 32*srcs*dests_total */
+    const uint64_t *__restrict__ src_ptrs,   /* [stripes][srcs]  */
+    const uint64_t *__restrict__ dst_ptrs,   /* [stripes][dests_total] */
+    int dests_total, uint32_t tiles_per_part, uint32_t total_tiles) {
+	constexpr uint32_t kTile = kChunkBytes * CH;
+	extern __shared__ __attribute__((aligned(16))) uint8_t smem[];
+	const uint32_t tid = threadIdx.x;
+
+	/* Stage this pass's table rows: [D][srcs][32] bytes. */
+	{
+		const uint8_t *src_tbl = gftbls_dev + (size_t)dest_base * srcs * 32;
+		int nbytes = D * srcs * 32;
+		for (int i = tid * 16; i < nbytes; i += kThreads * 16)
+			*(uint4 *)(smem + i) = *(const uint4 *)(src_tbl + i);
+	}
+	__syncthreads();
+
+	uint32_t b0 = SWZ ? xcd_remap(blockIdx.x, gridDim.x) : blockIdx.x;
+	for (uint32_t tile = b0; tile < total_tiles; tile += gridDim.x) {
+		uint32_t stripe = tile / tiles_per_part;
+		uint32_t tin = tile - stripe * tiles_per_part;
+		uint32_t base = tin * kTile + tid * 16u;
+		const uint64_t *sp_tab = src_ptrs + (uint64_t)stripe * srcs;
+		const uint64_t *dp_tab =
+		    dst_ptrs + (uint64_t)stripe * dests_total + dest_base;
+
+		uint4 acc[D][CH];
+#pragma unroll
+		for (int d = 0; d < D; ++d)
+#pragma unroll
+			for (int c = 0; c < CH; ++c)
+				acc[d][c] = make_uint4(0, 0, 0, 0);
+
+		if ((uint64_t)(tin + 1) * kTile <= part_len) {
+			/* full tile: branchless, next source prefetched while the
+			 * current one is accumulated */
+			uint4 w[CH], wn[CH];
+			{
+				const uint8_t *sp = (const uint8_t *)sp_tab[0];
+#pragma unroll
+				for (int c = 0; c < CH; ++c)
+					w[c] = *(const uint4 *)(sp + (base + c * kChunkBytes));
+			}
+			for (int j = 0; j < srcs; ++j) {
+				if (j + 1 < srcs) {
+					const uint8_t *spn = (const uint8_t *)sp_tab[j + 1];
+#pragma unroll
+					for (int c = 0; c < CH; ++c)
+						wn[c] = *(const uint4 *)(spn + (base + c * kChunkBytes));
+				}
+				uint4 L[D], H[D];
+#pragma unroll
+				for (int d = 0; d < D; ++d) {
+					const uint8_t *tb = smem + ((size_t)d * srcs + j) * 32;
+					L[d] = *(const uint4 *)tb;
+					H[d] = *(const uint4 *)(tb + 16);
+				}
+#pragma unroll
+				for (int c = 0; c < CH; ++c)
+					gf_macc_all<D, CH>(acc, c, w[c], L, H);
+#pragma unroll
+				for (int c = 0; c < CH; ++c) w[c] = wn[c];
+			}
+#pragma unroll
+			for (int d = 0; d < D; ++d) {
+				uint8_t *dp = (uint8_t *)dp_tab[d];
+#pragma unroll
+				for (int c = 0; c < CH; ++c) {
+					uint4 *p = (uint4 *)(dp + (base + c * kChunkBytes));
+					if (NTST) {
+						typedef unsigned int u32x4
+						    __attribute__((ext_vector_type(4)));
+						u32x4 v = {acc[d][c].x, acc[d][c].y, acc[d][c].z,
+						           acc[d][c].w};
+						__builtin_nontemporal_store(v, (u32x4 *)p);
+					} else {
+						*p = acc[d][c];
+					}
+				}
+			}
+		} else {
+			/* ragged tail tile: per-strip bounds checks */
+			for (int j = 0; j < srcs; ++j) {
+				const uint8_t *sp = (const uint8_t *)sp_tab[j];
+				uint4 L[D], H[D];
+#pragma unroll
+				for (int d = 0; d < D; ++d) {
+					const uint8_t *tb = smem + ((size_t)d * srcs + j) * 32;
+					L[d] = *(const uint4 *)tb;
+					H[d] = *(const uint4 *)(tb + 16);
+				}
+#pragma unroll
+				for (int c = 0; c < CH; ++c) {
+					uint32_t off = base + c * kChunkBytes;
+					if (off < part_len) {
+						uint4 w = *(const uint4 *)(sp + off);
+						gf_macc_all<D, CH>(acc, c, w, L, H);
+					}
+				}
+			}
+#pragma unroll
+			for (int d = 0; d < D; ++d) {
+				uint8_t *dp = (uint8_t *)dp_tab[d];
+#pragma unroll
+				for (int c = 0; c < CH; ++c) {
+					uint32_t off = base + c * kChunkBytes;
+					if (off < part_len) *(uint4 *)(dp + off) = acc[d][c];
+				}
+			}
+		}
+	}
+}

@@ -34,162 +34,15 @@
 		if (_e != hipSuccess) return LIZEC_EHIP;                         \
 	} while (0)
 
-/* ------------------------------------------------------------------ */
-/* EC encode/decode kernel                                            */
-/* ------------------------------------------------------------------ */
+#include "ec_kernel.h"
 
-/* Tile geometry: 256 threads/block; a tile is 16 KiB of each part, laid out
- * as 4 chunks of 4 KiB; in chunk c, thread t owns bytes
- * [c*4096 + t*16, +16) — so every wave-level load/store instruction touches
- * 64 consecutive 16-byte lanes = one coalesced 1 KiB segment. */
-constexpr int kThreads = 256;
-constexpr int kChunks = 4;
-constexpr uint32_t kChunkBytes = kThreads * 16;           /* 4096  */
-constexpr uint32_t kTileBytes = kChunkBytes * kChunks;    /* 16384 */
-
-/* 16-entry byte LUT on 4 packed nibbles via v_perm_b32.
- * t0..t3 = table bytes [0..3],[4..7],[8..11],[12..15]; nib = 4 nibble values
- * (one per byte, 0..15).  v_perm selector semantics: sel byte 0-3 picks a
- * byte of src1, 4-7 a byte of src0. */
-__device__ __forceinline__ uint32_t lut16(uint32_t t0, uint32_t t1,
-                                          uint32_t t2, uint32_t t3,
-                                          uint32_t nib) {
-	uint32_t s3 = nib & 0x07070707u;
-	uint32_t lo = __builtin_amdgcn_perm(t1, t0, s3);   /* nib in 0..7  */
-	uint32_t hi = __builtin_amdgcn_perm(t3, t2, s3);   /* nib in 8..15 */
-	uint32_t msel = 0x03020100u | ((nib >> 1) & 0x04040404u);
-	return __builtin_amdgcn_perm(hi, lo, msel);
-}
-
-/* GF(2^8) multiply-accumulate of one 32-bit word against one coefficient
- * table (L = products of low nibbles, H = of high nibbles). */
-__device__ __forceinline__ uint32_t gf_macc(uint32_t acc, uint32_t w,
-                                            const uint4 &L, const uint4 &H) {
-	uint32_t nl = w & 0x0f0f0f0fu;
-	uint32_t nh = (w >> 4) & 0x0f0f0f0fu;
-	return acc ^ lut16(L.x, L.y, L.z, L.w, nl) ^ lut16(H.x, H.y, H.z, H.w, nh);
-}
-
-template <int D>
-__device__ __forceinline__ void gf_macc_all(uint4 (&acc)[D][kChunks], int c,
-                                            const uint4 &w, const uint4 (&L)[D],
-                                            const uint4 (&H)[D]) {
-#pragma unroll
-	for (int d = 0; d < D; ++d) {
-		acc[d][c].x = gf_macc(acc[d][c].x, w.x, L[d], H[d]);
-		acc[d][c].y = gf_macc(acc[d][c].y, w.y, L[d], H[d]);
-		acc[d][c].z = gf_macc(acc[d][c].z, w.z, L[d], H[d]);
-		acc[d][c].w = gf_macc(acc[d][c].w, w.w, L[d], H[d]);
-	}
-}
-
-/* D = number of destination parts computed per pass (template keeps the
- * accumulators in VGPRs).  Each block walks tiles grid-stride; gftbls rows
- * [dest_base, dest_base+D) are staged once per block into LDS and re-read
- * as wave-uniform broadcasts once per (dest, src) per tile. */
-template <int D>
-__global__ __launch_bounds__(kThreads) void ec_encode_kernel(
-    uint32_t part_len, int srcs, int dest_base,
-    const uint8_t *__restrict__ gftbls_dev,  /* 32*srcs*dests_total */
-    const uint64_t *__restrict__ src_ptrs,   /* [stripes][srcs]  */
-    const uint64_t *__restrict__ dst_ptrs,   /* [stripes][dests_total] */
-    int dests_total, uint32_t tiles_per_part, uint32_t total_tiles) {
-	extern __shared__ __attribute__((aligned(16))) uint8_t smem[];
-	const uint32_t tid = threadIdx.x;
-
-	/* Stage this pass's table rows: [D][srcs][32] bytes. */
-	{
-		const uint8_t *src_tbl = gftbls_dev + (size_t)dest_base * srcs * 32;
-		int nbytes = D * srcs * 32;
-		for (int i = tid * 16; i < nbytes; i += kThreads * 16)
-			*(uint4 *)(smem + i) = *(const uint4 *)(src_tbl + i);
-	}
-	__syncthreads();
-
-	for (uint32_t tile = blockIdx.x; tile < total_tiles; tile += gridDim.x) {
-		uint32_t stripe = tile / tiles_per_part;
-		uint32_t tin = tile - stripe * tiles_per_part;
-		uint32_t base = tin * kTileBytes + tid * 16u;
-		const uint64_t *sp_tab = src_ptrs + (uint64_t)stripe * srcs;
-		const uint64_t *dp_tab =
-		    dst_ptrs + (uint64_t)stripe * dests_total + dest_base;
-
-		uint4 acc[D][kChunks];
-#pragma unroll
-		for (int d = 0; d < D; ++d)
-#pragma unroll
-			for (int c = 0; c < kChunks; ++c)
-				acc[d][c] = make_uint4(0, 0, 0, 0);
-
-		if ((tin + 1) * kTileBytes <= part_len) {
-			/* full tile: branchless, next source prefetched while the
-			 * current one is accumulated */
-			uint4 w[kChunks], wn[kChunks];
-			{
-				const uint8_t *sp = (const uint8_t *)sp_tab[0];
-#pragma unroll
-				for (int c = 0; c < kChunks; ++c)
-					w[c] = *(const uint4 *)(sp + (base + c * kChunkBytes));
-			}
-			for (int j = 0; j < srcs; ++j) {
-				if (j + 1 < srcs) {
-					const uint8_t *spn = (const uint8_t *)sp_tab[j + 1];
-#pragma unroll
-					for (int c = 0; c < kChunks; ++c)
-						wn[c] = *(const uint4 *)(spn + (base + c * kChunkBytes));
-				}
-				uint4 L[D], H[D];
-#pragma unroll
-				for (int d = 0; d < D; ++d) {
-					const uint8_t *tb = smem + ((size_t)d * srcs + j) * 32;
-					L[d] = *(const uint4 *)tb;
-					H[d] = *(const uint4 *)(tb + 16);
-				}
-#pragma unroll
-				for (int c = 0; c < kChunks; ++c)
-					gf_macc_all<D>(acc, c, w[c], L, H);
-#pragma unroll
-				for (int c = 0; c < kChunks; ++c) w[c] = wn[c];
-			}
-#pragma unroll
-			for (int d = 0; d < D; ++d) {
-				uint8_t *dp = (uint8_t *)dp_tab[d];
-#pragma unroll
-				for (int c = 0; c < kChunks; ++c)
-					*(uint4 *)(dp + (base + c * kChunkBytes)) = acc[d][c];
-			}
-		} else {
-			/* ragged tail tile: per-strip bounds checks */
-			for (int j = 0; j < srcs; ++j) {
-				const uint8_t *sp = (const uint8_t *)sp_tab[j];
-				uint4 L[D], H[D];
-#pragma unroll
-				for (int d = 0; d < D; ++d) {
-					const uint8_t *tb = smem + ((size_t)d * srcs + j) * 32;
-					L[d] = *(const uint4 *)tb;
-					H[d] = *(const uint4 *)(tb + 16);
-				}
-#pragma unroll
-				for (int c = 0; c < kChunks; ++c) {
-					uint32_t off = base + c * kChunkBytes;
-					if (off < part_len) {
-						uint4 w = *(const uint4 *)(sp + off);
-						gf_macc_all<D>(acc, c, w, L, H);
-					}
-				}
-			}
-#pragma unroll
-			for (int d = 0; d < D; ++d) {
-				uint8_t *dp = (uint8_t *)dp_tab[d];
-#pragma unroll
-				for (int c = 0; c < kChunks; ++c) {
-					uint32_t off = base + c * kChunkBytes;
-					if (off < part_len) *(uint4 *)(dp + off) = acc[d][c];
-				}
-			}
-		}
-	}
-}
+/* Product kernel configuration (chosen by the variant A/B harness,
+ * bench_variants.hip; see profiles/).  kChunks*4KiB = bytes of each part
+ * one 256-thread block covers per tile. */
+constexpr int kECChunks = 4;
+constexpr bool kECSwz = true;      /* +1.4% at 1024x64MiB (profiles/) */
+constexpr bool kECNtStore = true;
+constexpr uint32_t kTileBytes = kChunkBytes * kECChunks;
 
 /* ------------------------------------------------------------------ */
 /* CRC32 kernel                                                       */
@@ -423,7 +276,7 @@ static void launch_ec(uint32_t part_len, int srcs, int dest_base,
                       hipStream_t s) {
 	uint32_t grid = total_tiles < 262144u ? total_tiles : 262144u;
 	size_t lds = (size_t)D * srcs * 32;
-	hipLaunchKernelGGL(ec_encode_kernel<D>, dim3(grid), dim3(kThreads), lds, s,
+	hipLaunchKernelGGL(HIP_KERNEL_NAME(ec_encode_kernel<D, kECChunks, kECSwz, kECNtStore>), dim3(grid), dim3(kThreads), lds, s,
 	                   part_len, srcs, dest_base, d_tbls, d_src, d_dst,
 	                   dests_total, tiles_per_part, total_tiles);
 }

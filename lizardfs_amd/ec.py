@@ -99,6 +99,16 @@ class ReedSolomon:
         if t.dtype != torch.uint8 or not t.is_cuda or not t.is_contiguous():
             raise ValueError(f"{what} must be a contiguous CUDA uint8 tensor")
 
+    @staticmethod
+    def _frag_geom(t, what):
+        """[S, L] fragment, possibly a strided view (e.g. data[:, i, :]):
+        rows must be contiguous; returns (S, L, row_stride_bytes)."""
+        if t.dtype != torch.uint8 or not t.is_cuda or t.dim() != 2 or \
+                t.stride(1) != 1:
+            raise ValueError(f"{what} must be a CUDA uint8 [S, L] tensor "
+                             f"with contiguous rows")
+        return t.shape[0], t.shape[1], t.stride(0)
+
     def encode_batch(self, data, parity=None):
         """Compute parity for a batch of stripes.
 
@@ -113,6 +123,7 @@ class ReedSolomon:
             raise ValueError(f"data has {k} parts, expected {self.k}")
         if plen % 16:
             raise ValueError("part length must be a multiple of 16")
+        caller_parity = parity is not None
         if parity is None:
             parity = torch.empty((S, self.m, plen), dtype=torch.uint8,
                                  device=data.device)
@@ -128,11 +139,13 @@ class ReedSolomon:
                np.arange(S * k, dtype=np.uint64) * np.uint64(plen))
         dst = (parity.data_ptr() +
                np.arange(S * self.m, dtype=np.uint64) * np.uint64(plen))
-        plan_key = ("enc", data.data_ptr(), parity.data_ptr(), S, plen)
+        # plans are cached only for caller-owned (stable) output buffers
+        plan_key = (("enc", data.data_ptr(), parity.data_ptr(), S, plen)
+                    if caller_parity else None)
         self._run(plen, tbl, ic, oc, src, dst, S, plan_key=plan_key)
         return parity
 
-    def recover_batch(self, fragments, erased, want=None):
+    def recover_batch(self, fragments, erased, want=None, out=None):
         """Recover missing parts for a batch of stripes.
 
         fragments: list of k+m entries; entry i is a uint8 CUDA tensor [S, L]
@@ -164,9 +177,9 @@ class ReedSolomon:
                 continue
             present |= 1 << i
             if fragments[i] is not None:
-                self._check_part(fragments[i], f"fragment {i}")
+                s, l, _ = self._frag_geom(fragments[i], f"fragment {i}")
                 nonnull |= 1 << i
-                S, plen = fragments[i].shape
+                S, plen = s, l
                 dev = fragments[i].device
         for i in want:
             needed |= 1 << i
@@ -179,17 +192,29 @@ class ReedSolomon:
 
         srcs = [fragments[i] for i in range(nparts)
                 if (nonnull >> i) & 1]
-        outs = {i: torch.empty((S, plen), dtype=torch.uint8, device=dev)
-                for i in sorted(want)}
-        stride = np.arange(S, dtype=np.uint64) * np.uint64(plen)
+        if out is not None:
+            outs = out   # caller-provided stable buffers -> plan is cached
+            for i in want:
+                self._check_part(outs[i], f"out[{i}]")
+        else:
+            outs = {i: torch.empty((S, plen), dtype=torch.uint8, device=dev)
+                    for i in sorted(want)}
+        rows = np.arange(S, dtype=np.uint64)
         src = np.empty((S, ic), np.uint64)
         for j, t in enumerate(srcs):
-            src[:, j] = t.data_ptr() + stride
+            _, _, rstride = self._frag_geom(t, "fragment")
+            src[:, j] = t.data_ptr() + rows * np.uint64(rstride)
         dst = np.empty((S, oc), np.uint64)
         for j, i in enumerate(sorted(want)):
-            dst[:, j] = outs[i].data_ptr() + stride
+            dst[:, j] = outs[i].data_ptr() + rows * np.uint64(plen)
+        plan_key = None
+        if out is not None:
+            plan_key = ("rec", present, nonnull, needed,
+                        tuple(t.data_ptr() for t in srcs),
+                        tuple(outs[i].data_ptr() for i in sorted(want)),
+                        S, plen)
         self._run(plen, tbl, ic, oc, np.ascontiguousarray(src.ravel()),
-                  np.ascontiguousarray(dst.ravel()), S)
+                  np.ascontiguousarray(dst.ravel()), S, plan_key=plan_key)
         return outs
 
     def sync(self):

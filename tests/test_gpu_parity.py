@@ -130,6 +130,27 @@ def test_roundtrip_full_size_stripes(rs_mod):
         assert np.array_equal(got[l], exp[l])
 
 
+def test_recover_from_strided_views_with_out(rs_mod):
+    """Fragments may be strided views into the [S,k,L] batch (no copies),
+    and caller-provided outputs enable the cached-plan path."""
+    k, m, S, plen = 8, 2, 5, 32768
+    rng = np.random.default_rng(2024)
+    data_np = rng.integers(0, 256, (S, k, plen), np.uint8)
+    data = to_gpu(data_np)
+    rs = rs_mod(k, m)
+    parity = rs.encode_batch(data)
+    frags = [None if i in (1, 5) else
+             (data[:, i, :] if i < k else parity[:, i - k, :])
+             for i in range(k + m)]
+    outs = {i: torch.empty((S, plen), dtype=torch.uint8, device="cuda")
+            for i in (1, 5)}
+    for _ in range(3):   # repeated runs exercise the plan cache
+        rec = rs.recover_batch(frags, erased=(1, 5), out=outs)
+    rs.sync()
+    for i in (1, 5):
+        assert np.array_equal(rec[i].cpu().numpy(), data_np[:, i, :])
+
+
 def test_recover_with_null_zero_parts(rs_mod):
     """NULL surviving part = implicit zeros (reed_solomon.h:79) on GPU."""
     k, m, S, plen = 6, 2, 3, 4096
