@@ -12,7 +12,7 @@
  *    software-pipelined path (next source's strips prefetched while the
  *    current one is accumulated), ragged tails take a guarded path;
  *  - per-64KiB-block CRC32 (hddspacemgr.cc:1918 gate) with one wave per
- *    block: 64 lanes x 1 KiB segments, slicing-by-4 LDS tables, then a
+ *    block: 64 lanes x 1 KiB segments, slicing-by-8 LDS tables, then a
  *    shfl-based combine tree folding segment CRCs with the same GF(2)
  *    "advance by N zero bytes" matrices as mycrc32_combine (crc.cc:153-224).
  *
