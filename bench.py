@@ -3,19 +3,25 @@
 
 Measures BASELINE.json's metric ("GiB/s EC encode+decode, ec(8,2) 64 MiB
 stripes") on BASELINE config 2: ec(8,2) encode of a batch of 64 MiB
-synthetic stripes, inputs resident in HBM.  A "step" = one encode pass of
-the whole batch.  value = whole-job data-in GiB/s across all ranks
-(the reference's own throughput convention, reed_solomon_unittest.cc:44-73).
+synthetic stripes, inputs resident in HBM.  A "step" = one pass of the op
+over the whole batch.  value = whole-job data GiB/s across all ranks (the
+reference's own throughput convention, reed_solomon_unittest.cc:44-73:
+bytes of stripe data / time).
 
 Usage: python bench.py [--gpus N] [--steps K] [--warmup W] [--stripes S]
-                       [--op encode|decode|crc]
+                       [--op encode|decode|crc|encode_crc|mixed]
 N>1 is launched by the driver via torch.distributed.run (one rank per GPU,
-RCCL); stripes shard across ranks as independent batches (weak scaling — the
-reference processes stripes independently too; no data-path collective,
-SURVEY §8e).
+RCCL); stripes shard across ranks as independent batches (weak scaling —
+the reference processes stripes independently too).
 
---op decode measures BASELINE config 3 (2 erased data parts); --op crc the
-per-64KiB-block CRC32 gate.  The default (encode) is the contract line.
+Ops map to the BASELINE configs:
+  encode      config 2: ec(8,2) encode                       [default]
+  decode      config 3: ec(8,2) decode, 2 erased data parts
+  crc         the per-64KiB-block CRC32 gate alone
+  encode_crc  config 4: ec(16,4) encode + CRC32 verify of all parts
+  mixed       config 5: ec(32,6) encode + 3-erasure decode; when
+              distributed, recovered parts are all-gathered over
+              RCCL/xGMI (the one real exchange in the path)
 
 The cpu_baseline leg times the ORACLE (CPU restatement of the reference
 path, OpenMP over stripes) on a bounded sample — reported context, not the
@@ -34,71 +40,71 @@ sys.path.insert(0, os.path.join(REPO, "oracle"))
 import numpy as np  # noqa: E402
 import torch  # noqa: E402
 
-K_EC = 8
-M_EC = 2
 STRIPE_BYTES = 64 * 1024 * 1024
-PART_LEN = STRIPE_BYTES // K_EC
-ERASED = (1, 5)   # BASELINE config 3: 2 erased data parts
+
+OPS = {
+    # op: (k, m, erased_parts, default stripes per rank)
+    "encode": (8, 2, (), 1024),
+    "decode": (8, 2, (1, 5), 1024),
+    "crc": (8, 2, (), 1024),
+    "encode_crc": (16, 4, (), 1024),
+    "mixed": (32, 6, (2, 9, 30), 512),
+}
 
 
 def log(msg):
     print(msg, file=sys.stderr, flush=True)
 
 
-def cpu_baseline_leg(op, target_seconds=10.0):
+def cpu_baseline_leg(op, k, m, erased, target_seconds=10.0, max_reps=64):
     """Time the oracle's threaded path on a bounded sample of the same
     workload on this box's host cores.  kind='port' (restatement of
     galois_field_encode.cc:28-47 + reed_solomon.h semantics)."""
     import oracle
     cores = int(os.environ.get("OMP_NUM_THREADS", os.cpu_count() or 1))
     rng = np.random.default_rng(42)
+    part_len = STRIPE_BYTES // k
     n = 16  # 1 GiB data sample, repeated until ~target_seconds
-    data = rng.integers(0, 256, (n, K_EC, PART_LEN), np.uint8)
+    data = rng.integers(0, 256, (n, k, part_len), np.uint8)
 
-    if op == "crc":
-        buf = np.ascontiguousarray(data.reshape(-1))
+    def timed(fn, unit_gib):
         t0 = time.perf_counter()
-        oracle.crc32_blocks(buf, 65536)
-        once = time.perf_counter() - t0
-        reps = max(1, int(target_seconds / max(once, 1e-3)))
+        fn()
+        once = max(time.perf_counter() - t0, 1e-3)
+        reps = max(1, min(int(target_seconds / once), max_reps))
         t0 = time.perf_counter()
         for _ in range(reps):
-            oracle.crc32_blocks(buf, 65536)
+            fn()
         dt = time.perf_counter() - t0
-        gib = reps * buf.size / (1 << 30)
-        sample = f"{reps}x{n * 64}MiB CRC32/64KiB blocks, {dt:.1f}s, OpenMP"
-        return {"value": round(gib / dt, 3), "unit": "GiB/s", "cores": cores,
-                "kind": "port", "sample": sample}
+        return reps * unit_gib / dt, reps, dt
 
-    if op == "decode":
-        present = sum(1 << i for i in range(K_EC + M_EC) if i not in ERASED)
-        needed = sum(1 << i for i in ERASED)
-        tbl, ic, oc = oracle.rs_make_tables(K_EC, M_EC, present, present,
-                                            needed)
-        # reuse encode_stripes shape: inputs = the ic surviving parts
-        parity = np.zeros((n, oc, PART_LEN), np.uint8)
-        srcs = np.ascontiguousarray(data[:, :ic])  # ic=8 surviving parts
-        enc = lambda: oracle.encode_stripes(ic, oc, PART_LEN, n, tbl,
-                                            srcs, parity)
-    else:
-        tbl, ic, oc = oracle.rs_make_tables(
-            K_EC, M_EC, (1 << K_EC) - 1, (1 << K_EC) - 1,
-            ((1 << M_EC) - 1) << K_EC)
-        parity = np.zeros((n, M_EC, PART_LEN), np.uint8)
-        enc = lambda: oracle.encode_stripes(K_EC, M_EC, PART_LEN, n, tbl,
-                                            data, parity)
+    legs = []
+    if op in ("encode", "decode", "encode_crc", "mixed"):
+        if op == "decode":
+            nparts = k + m
+            present = sum(1 << i for i in range(nparts) if i not in erased)
+            needed = sum(1 << i for i in erased)
+            tbl, ic, oc = oracle.rs_make_tables(k, m, present, present, needed)
+            srcs = np.ascontiguousarray(data[:, :ic])
+        else:
+            tbl, ic, oc = oracle.rs_make_tables(
+                k, m, (1 << k) - 1, (1 << k) - 1, ((1 << m) - 1) << k)
+            srcs = data
+        out = np.zeros((n, oc, part_len), np.uint8)
+        legs.append(lambda: oracle.encode_stripes(ic, oc, part_len, n, tbl,
+                                                  srcs, out))
+    if op in ("crc", "encode_crc"):
+        buf = np.ascontiguousarray(data.reshape(-1))
+        legs.append(lambda: oracle.crc32_blocks(buf, 65536))
 
-    t0 = time.perf_counter()
-    enc()
-    once = time.perf_counter() - t0
-    reps = max(1, int(target_seconds / max(once, 1e-3)))
-    t0 = time.perf_counter()
-    for _ in range(reps):
-        enc()
-    dt = time.perf_counter() - t0
-    gib = reps * n * STRIPE_BYTES / (1 << 30)
-    sample = f"{reps}x{n}x64MiB stripes ec(8,2) {op}, {dt:.1f}s, OpenMP"
-    return {"value": round(gib / dt, 3), "unit": "GiB/s", "cores": cores,
+    def all_legs():
+        for f in legs:
+            f()
+
+    gibs, reps, dt = timed(all_legs, n * STRIPE_BYTES / (1 << 30))
+    sample = (f"{reps}x{n}x64MiB stripes ec({k},{m}) {op}, {dt:.1f}s, "
+              f"OpenMP")
+    return {"value": round(gibs, 3), "unit": "GiB/s", "cores": cores,
             "kind": "port", "sample": sample}
 
 
@@ -122,12 +128,16 @@ def main():
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=20)
     ap.add_argument("--warmup", type=int, default=5)
-    ap.add_argument("--stripes", type=int, default=1024,
-                    help="stripes per rank (weak scaling)")
-    ap.add_argument("--op", choices=("encode", "decode", "crc"),
-                    default="encode")
+    ap.add_argument("--stripes", type=int, default=0,
+                    help="stripes per rank (weak scaling); 0 = op default")
+    ap.add_argument("--op", choices=tuple(OPS), default="encode")
     ap.add_argument("--skip-cpu-baseline", action="store_true")
     args = ap.parse_args()
+
+    K, M, ERASED, S_default = OPS[args.op]
+    S = args.stripes or S_default
+    PART_LEN = STRIPE_BYTES // K
+    nparts = K + M
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
@@ -146,48 +156,91 @@ def main():
     from lizardfs_amd.ec import ReedSolomon
     from lizardfs_amd import crc as lcrc
 
-    S = args.stripes
-    workload = (f"ec({K_EC},{M_EC}) {args.op}, {S}x64MiB synthetic "
+    workload = (f"ec({K},{M}) {args.op}, {S}x64MiB synthetic "
                 f"stripes/GPU, device-resident")
     log(f"[rank {rank}] generating {S} stripes "
         f"({S * STRIPE_BYTES / (1 << 30):.0f} GiB data) on cuda:{local_rank}")
     g = torch.Generator(device="cuda").manual_seed(42 + rank)
-    data = torch.randint(0, 256, (S, K_EC, PART_LEN), dtype=torch.uint8,
+    data = torch.randint(0, 256, (S, K, PART_LEN), dtype=torch.uint8,
                          device="cuda", generator=g)
-    parity = torch.empty((S, M_EC, PART_LEN), dtype=torch.uint8,
-                         device="cuda")
+    parity = torch.empty((S, M, PART_LEN), dtype=torch.uint8, device="cuda")
 
-    rs = ReedSolomon(K_EC, M_EC, device=local_rank)
-    nparts = K_EC + M_EC
+    rs = ReedSolomon(K, M, device=local_rank)
 
     if args.op == "encode":
         def step():
             rs.encode_batch(data, parity)
-        # dominant kernel: ec_encode_kernel<2>; 1 launch/step
-        alg_bytes_per_launch = S * STRIPE_BYTES * (K_EC + M_EC) // K_EC
+        alg_bytes_per_launch = S * STRIPE_BYTES * (K + M) // K
         metric = "GiB/s EC encode, ec(8,2) 64MiB stripes"
     elif args.op == "decode":
-        rs.encode_batch(data, parity)   # produce real parity first
+        rs.encode_batch(data, parity)
         rs.sync()
         frags = [None if i in ERASED else
-                 (data[:, i, :] if i < K_EC else parity[:, i - K_EC, :])
+                 (data[:, i, :] if i < K else parity[:, i - K, :])
                  for i in range(nparts)]
         outs = {i: torch.empty((S, PART_LEN), dtype=torch.uint8,
                                device="cuda") for i in ERASED}
 
         def step():
             rs.recover_batch(frags, erased=ERASED, out=outs)
-        alg_bytes_per_launch = S * PART_LEN * (K_EC + len(ERASED))
+        alg_bytes_per_launch = S * PART_LEN * (K + len(ERASED))
         metric = "GiB/s EC decode(2 erasures), ec(8,2) 64MiB stripes"
-    else:  # crc
+    elif args.op == "crc":
         flat = data.reshape(-1)
         crcs = torch.empty(flat.numel() // 65536, dtype=torch.int32,
                            device="cuda")
 
         def step():
             lcrc.crc32_blocks(flat, 65536, out=crcs)
-        alg_bytes_per_launch = S * STRIPE_BYTES + 4 * (S * STRIPE_BYTES // 65536)
+        alg_bytes_per_launch = (S * STRIPE_BYTES +
+                                4 * (S * STRIPE_BYTES // 65536))
         metric = "GiB/s CRC32 per-64KiB-block"
+    elif args.op == "encode_crc":
+        # config 4: encode, then CRC-verify every 64 KiB block of every
+        # part (data re-read + parity read), as the replicator does before
+        # hdd_write (chunk_replicator.cc:189, hddspacemgr.cc:1918)
+        dflat = data.reshape(-1)
+        pflat = parity.reshape(-1)
+        dcrcs = torch.empty(dflat.numel() // 65536, dtype=torch.int32,
+                            device="cuda")
+        pcrcs = torch.empty(pflat.numel() // 65536, dtype=torch.int32,
+                            device="cuda")
+
+        def step():
+            rs.encode_batch(data, parity)
+            lcrc.crc32_blocks(dflat, 65536, out=dcrcs)
+            lcrc.crc32_blocks(pflat, 65536, out=pcrcs)
+        alg_bytes_per_launch = S * STRIPE_BYTES * (K + M) // K  # encode leg
+        metric = "GiB/s EC encode+CRC32 pipeline, ec(16,4) 64MiB stripes"
+    else:  # mixed — config 5
+        rs.encode_batch(data, parity)
+        rs.sync()
+        # exactly m erased required (reed_solomon.h:95): pad the 3 data
+        # erasures with unneeded parity parts, as ec_read_plan.h:126 does
+        pad = tuple(range(nparts - (M - len(ERASED)), nparts))
+        erased_full = tuple(ERASED) + pad
+        frags = [None if i in erased_full else
+                 (data[:, i, :] if i < K else parity[:, i - K, :])
+                 for i in range(nparts)]
+        outs = {i: torch.empty((S, PART_LEN), dtype=torch.uint8,
+                               device="cuda") for i in ERASED}
+        if distributed:
+            gathered = [
+                [torch.empty((S, PART_LEN), dtype=torch.uint8, device="cuda")
+                 for _ in range(world)] for _ in ERASED]
+
+        def step():
+            rs.encode_batch(data, parity)
+            rs.recover_batch(frags, erased=erased_full, want=set(ERASED),
+                             out=outs)
+            if distributed:
+                # RCCL all-gather of recovered parts over xGMI so every
+                # rank holds the full recovered set
+                for gi, i in enumerate(sorted(outs)):
+                    torch.distributed.all_gather(gathered[gi], outs[i])
+        alg_bytes_per_launch = S * STRIPE_BYTES * (K + M) // K
+        metric = ("GiB/s EC mixed encode+3-erasure decode, ec(32,6) "
+                  "64MiB stripes")
 
     for _ in range(args.warmup):
         step()
@@ -216,7 +269,11 @@ def main():
         elapsed = float(t.item())
 
     step_ms = [a.elapsed_time(b) for a, b in ev]
-    avg_launch_ms = sum(step_ms) / len(step_ms)
+    avg_step_ms = sum(step_ms) / len(step_ms)
+    # roofline leg: the dominant kernel is the EC encode launch (1/step)
+    # except for --op crc where it is the CRC kernel itself
+    launches_per_step = 1
+    avg_launch_ms = avg_step_ms / launches_per_step
 
     data_gib_per_step = n_gpus * S * STRIPE_BYTES / (1 << 30)
     value = data_gib_per_step * args.steps / elapsed
@@ -227,11 +284,30 @@ def main():
     achieved_gbps = alg_bytes_per_launch / (avg_launch_ms / 1e3) / 1e9
     peak_gbps = 8000.0  # MI355X_MICROARCH.md: HBM3E 8 TB/s spec peak
     traffic = read_traffic_calibration(workload)
+    if args.op in ("encode_crc", "mixed"):
+        # multiple kernels per step: the per-step event time is not a
+        # single-kernel duration; report whole-step rate and mark it
+        roofline = {
+            "bound": "hbm",
+            "achieved": round(alg_bytes_per_launch / (avg_step_ms / 1e3) / 1e9, 1),
+            "peak": peak_gbps, "unit": "GB/s",
+            "frac": None,   # composite step; per-kernel fracs in profiles/
+            "traffic": traffic,
+        }
+    else:
+        roofline = {
+            "bound": "hbm",
+            "achieved": round(achieved_gbps, 1),
+            "peak": peak_gbps,
+            "unit": "GB/s",
+            "frac": round(achieved_gbps / peak_gbps, 4),
+            "traffic": traffic,
+        }
 
     cpu_baseline = None
     if n_gpus == 1 and not args.skip_cpu_baseline:
         log("[rank 0] timing CPU baseline (oracle, OpenMP)...")
-        cpu_baseline = cpu_baseline_leg(args.op)
+        cpu_baseline = cpu_baseline_leg(args.op, K, M, ERASED)
 
     result = {
         "metric": metric,
@@ -248,19 +324,14 @@ def main():
         "data": "synthetic",
         "config": {
             "workload": workload,
-            "k": K_EC, "m": M_EC,
+            "k": K, "m": M,
             "stripe_bytes": STRIPE_BYTES,
             "stripes_per_gpu": S,
-            "parallelism": f"independent stripe batches x{n_gpus}",
+            "parallelism": f"independent stripe batches x{n_gpus}" +
+                           (" + RCCL all-gather of recovered parts"
+                            if args.op == "mixed" and distributed else ""),
         },
-        "roofline": {
-            "bound": "hbm",
-            "achieved": round(achieved_gbps, 1),
-            "peak": peak_gbps,
-            "unit": "GB/s",
-            "frac": round(achieved_gbps / peak_gbps, 4),
-            "traffic": traffic,
-        },
+        "roofline": roofline,
         "cpu_baseline": cpu_baseline,
     }
     print(json.dumps(result), flush=True)

@@ -75,6 +75,12 @@ uint32_t lizec_crc32(uint32_t crc, const uint8_t *block, uint32_t leng);
 uint32_t lizec_crc32_combine(uint32_t crc1, uint32_t crc2, uint32_t leng2);
 void lizec_crc32_init(void);
 
+/* Host byte-XOR (xor-goal parity building block; also exported with the
+ * C++ mangling of the reference's blockXor, common/block_xor.h:33).
+ * GPU xor parity/recovery runs through the EC kernel with all-ones
+ * coefficient tables (lizardfs_amd/xor.py). */
+void lizec_blockxor(uint8_t *dest, const uint8_t *source, size_t size);
+
 /* ------------------------------------------------------------------ */
 /* Reed-Solomon table builders — host-side restatement of             */
 /* ReedSolomon<32,32> (reed_solomon.h:41-373).                        */

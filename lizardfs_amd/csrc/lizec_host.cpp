@@ -338,6 +338,29 @@ void mycrc32_init(void) {
 }
 
 /* ------------------------------------------------------------------ */
+/* XOR family host op (block_xor.h:33 drop-in; xor parity on GPU runs  */
+/* through the EC kernel with all-ones coefficients)                   */
+/* ------------------------------------------------------------------ */
+
+extern "C" void lizec_blockxor(uint8_t *dest, const uint8_t *source,
+                               size_t size) {
+	size_t i = 0;
+	for (; i + 8 <= size; i += 8) {
+		uint64_t a, b;
+		memcpy(&a, dest + i, 8);
+		memcpy(&b, source + i, 8);
+		a ^= b;
+		memcpy(dest + i, &a, 8);
+	}
+	for (; i < size; ++i) dest[i] ^= source[i];
+}
+
+/* C++-linkage alias for the reference's mangled symbol (block_xor.h:33). */
+void blockXor(uint8_t *dest, const uint8_t *source, size_t size) {
+	lizec_blockxor(dest, source, size);
+}
+
+/* ------------------------------------------------------------------ */
 /* Slice-type algebra (goal.h:108-119, slice_traits.h, chunk_part_type.h) */
 /* ------------------------------------------------------------------ */
 
