@@ -54,16 +54,21 @@ static void launch_var(uint32_t part_len, int srcs, const uint8_t *tbls,
                        uint32_t grid_cap, hipStream_t s) {
 	uint32_t grid = total_tiles < grid_cap ? total_tiles : grid_cap;
 	size_t lds = (size_t)D * srcs * 32;
-	hipLaunchKernelGGL(HIP_KERNEL_NAME(ec_encode_kernel<D, CH, SWZ, NTST>),
-	                   dim3(grid), dim3(kThreads), lds, s, part_len, srcs, 0,
-	                   tbls, src, dst, dests, tiles_per_part, total_tiles);
+	for (int base = 0; base + D <= dests; base += D)
+		hipLaunchKernelGGL(HIP_KERNEL_NAME(ec_encode_kernel<D, CH, SWZ, NTST>),
+		                   dim3(grid), dim3(kThreads), lds, s, part_len, srcs,
+		                   base, tbls, src, dst, dests, tiles_per_part,
+		                   total_tiles);
 }
 
 int main(int argc, char **argv) {
 	int k = 8, m = 2, stripes = 1024, reps = 6;
-	uint64_t part_len = 8u * 1024 * 1024;
-	if (argc > 1) stripes = atoi(argv[1]);
-	if (argc > 2) reps = atoi(argv[2]);
+	if (argc > 1) k = atoi(argv[1]);
+	if (argc > 2) m = atoi(argv[2]);
+	if (argc > 3) stripes = atoi(argv[3]);
+	if (argc > 4) reps = atoi(argv[4]);
+	uint64_t part_len = 64ull * 1024 * 1024 / k;
+	part_len &= ~15ull;
 
 	size_t data_bytes = (size_t)stripes * k * part_len;
 	size_t par_bytes = (size_t)stripes * m * part_len;
@@ -76,7 +81,7 @@ int main(int argc, char **argv) {
 	hipLaunchKernelGGL(fill_kernel, dim3(4096), dim3(256), 0, 0, d_data,
 	                   data_bytes, 0x1234567u);
 
-	uint8_t tbls[32 * 32 * 2];
+	static uint8_t tbls[32 * 32 * 32];
 	if (lizec_rs_encode_tables(k, m, tbls) != 0) return 1;
 	uint8_t *d_tbls;
 	CK(hipMalloc(&d_tbls, 32 * k * m));
@@ -114,19 +119,21 @@ int main(int argc, char **argv) {
 		           const uint64_t *, int, uint32_t, uint32_t, uint32_t,
 		           hipStream_t);
 		int ch;
+		int d;
 		uint32_t grid_cap;
 	};
 	Cfg cfgs[] = {
-	    {"D2_CH4_base      ", launch_var<2, 4, false, false>, 4, 262144},
-	    {"D2_CH4_swz       ", launch_var<2, 4, true, false>, 4, 262144},
-	    {"D2_CH4_nt        ", launch_var<2, 4, false, true>, 4, 262144},
-	    {"D2_CH4_swz_nt    ", launch_var<2, 4, true, true>, 4, 262144},
-	    {"D2_CH8_base      ", launch_var<2, 8, false, false>, 8, 262144},
-	    {"D2_CH8_swz_nt    ", launch_var<2, 8, true, true>, 8, 262144},
-	    {"D2_CH2_base      ", launch_var<2, 2, false, false>, 2, 262144},
-	    {"D2_CH4_grid16k   ", launch_var<2, 4, false, false>, 4, 16384},
-	    {"D2_CH4_swz_g16k  ", launch_var<2, 4, true, false>, 4, 16384},
-	    {"D2_CH4_nt_g2048  ", launch_var<2, 4, false, true>, 4, 2048},
+	    {"D1_CH4_swz_nt ", launch_var<1, 4, true, true>, 4, 1, 262144},
+	    {"D2_CH4_swz_nt ", launch_var<2, 4, true, true>, 4, 2, 262144},
+	    {"D2_CH4_base   ", launch_var<2, 4, false, false>, 4, 2, 262144},
+	    {"D3_CH4_swz_nt ", launch_var<3, 4, true, true>, 4, 3, 262144},
+	    {"D4_CH4_swz_nt ", launch_var<4, 4, true, true>, 4, 4, 262144},
+	    {"D5_CH2_swz_nt ", launch_var<5, 2, true, true>, 2, 5, 262144},
+	    {"D6_CH4_swz_nt ", launch_var<6, 4, true, true>, 4, 6, 262144},
+	    {"D6_CH3_swz_nt ", launch_var<6, 3, true, true>, 3, 6, 262144},
+	    {"D6_CH2_swz_nt ", launch_var<6, 2, true, true>, 2, 6, 262144},
+	    {"D8_CH2_swz_nt ", launch_var<8, 2, true, true>, 2, 8, 262144},
+	    {"D8_CH1_swz_nt ", launch_var<8, 1, true, true>, 1, 8, 262144},
 	};
 
 	double traffic = (double)stripes * part_len * (k + m);
@@ -135,6 +142,7 @@ int main(int argc, char **argv) {
 	CK(hipEventCreate(&e1));
 
 	for (auto &c : cfgs) {
+		if (m % c.d != 0) { printf("%s  skip (D !| m)\n", c.name); continue; }
 		uint32_t tpp = (uint32_t)((part_len + c.ch * kChunkBytes - 1) /
 		                          (c.ch * kChunkBytes));
 		uint32_t tot = tpp * stripes;

@@ -360,9 +360,12 @@ static int run_batch(uint64_t part_len, int srcs, int dests,
                      const uint8_t *d_tbls, const uint64_t *d_src,
                      const uint64_t *d_dst, uint32_t tiles_per_part,
                      uint32_t total_tiles, hipStream_t s) {
+	/* One pass computes up to kMaxDestsPerPass destinations; wider m splits
+	 * into passes (each pass re-reads the sources, so fewer passes wins on
+	 * this HBM-bound kernel — D<=8 measured best, profiles/). */
 	for (int base = 0; base < dests;) {
 		int d = dests - base;
-		if (d > 4) d = 4;
+		if (d > 8) d = 8;
 		switch (d) {
 		case 1: launch_ec<1>((uint32_t)part_len, srcs, base, d_tbls, d_src,
 		                     d_dst, dests, tiles_per_part, total_tiles, s);
@@ -373,7 +376,19 @@ static int run_batch(uint64_t part_len, int srcs, int dests,
 		case 3: launch_ec<3>((uint32_t)part_len, srcs, base, d_tbls, d_src,
 		                     d_dst, dests, tiles_per_part, total_tiles, s);
 			break;
-		default: launch_ec<4>((uint32_t)part_len, srcs, base, d_tbls, d_src,
+		case 4: launch_ec<4>((uint32_t)part_len, srcs, base, d_tbls, d_src,
+		                     d_dst, dests, tiles_per_part, total_tiles, s);
+			break;
+		case 5: launch_ec<5>((uint32_t)part_len, srcs, base, d_tbls, d_src,
+		                     d_dst, dests, tiles_per_part, total_tiles, s);
+			break;
+		case 6: launch_ec<6>((uint32_t)part_len, srcs, base, d_tbls, d_src,
+		                     d_dst, dests, tiles_per_part, total_tiles, s);
+			break;
+		case 7: launch_ec<7>((uint32_t)part_len, srcs, base, d_tbls, d_src,
+		                     d_dst, dests, tiles_per_part, total_tiles, s);
+			break;
+		default: launch_ec<8>((uint32_t)part_len, srcs, base, d_tbls, d_src,
 		                      d_dst, dests, tiles_per_part, total_tiles, s);
 			break;
 		}
