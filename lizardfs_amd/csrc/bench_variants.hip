@@ -128,6 +128,8 @@ int main(int argc, char **argv) {
 	    {"D2_CH4_base   ", launch_var<2, 4, false, false>, 4, 2, 262144},
 	    {"D3_CH4_swz_nt ", launch_var<3, 4, true, true>, 4, 3, 262144},
 	    {"D4_CH4_swz_nt ", launch_var<4, 4, true, true>, 4, 4, 262144},
+	    {"D4_CH3_swz_nt ", launch_var<4, 3, true, true>, 3, 4, 262144},
+	    {"D4_CH2_swz_nt ", launch_var<4, 2, true, true>, 2, 4, 262144},
 	    {"D5_CH2_swz_nt ", launch_var<5, 2, true, true>, 2, 5, 262144},
 	    {"D6_CH4_swz_nt ", launch_var<6, 4, true, true>, 4, 6, 262144},
 	    {"D6_CH3_swz_nt ", launch_var<6, 3, true, true>, 3, 6, 262144},

@@ -37,12 +37,14 @@
 #include "ec_kernel.h"
 
 /* Product kernel configuration (chosen by the variant A/B harness,
- * bench_variants.hip; see profiles/).  kChunks*4KiB = bytes of each part
- * one 256-thread block covers per tile. */
-constexpr int kECChunks = 4;
-constexpr bool kECSwz = true;      /* +1.4% at 1024x64MiB (profiles/) */
+ * bench_variants.hip; numbers in profiles/ROUND1.md):
+ *  - XCD-bijective block remap + non-temporal parity stores everywhere;
+ *  - tile size per destination-group width D: D<=2 uses 4-chunk (16 KiB)
+ *    tiles; D>=3 uses 2-chunk (8 KiB) tiles (fewer accumulator VGPRs ->
+ *    4 waves/SIMD; ec(16,4) D4: 3953 -> 4113 GB/s, ec(32,6) D6: 3001). */
+constexpr bool kECSwz = true;
 constexpr bool kECNtStore = true;
-constexpr uint32_t kTileBytes = kChunkBytes * kECChunks;
+constexpr int ec_chunks_for(int d) { return d <= 2 ? 4 : 2; }
 
 /* ------------------------------------------------------------------ */
 /* CRC32 kernel                                                       */
@@ -228,8 +230,6 @@ struct lizec_plan {
 	lizec_engine *e;
 	uint64_t part_len;
 	int srcs, dests, num_stripes;
-	uint32_t tiles_per_part;
-	uint32_t total_tiles;
 	uint8_t *d_tbls;
 	uint64_t *d_src;
 	uint64_t *d_dst;
@@ -347,19 +347,25 @@ template <int D>
 static void launch_ec(uint32_t part_len, int srcs, int dest_base,
                       const uint8_t *d_tbls, const uint64_t *d_src,
                       const uint64_t *d_dst, int dests_total,
-                      uint32_t tiles_per_part, uint32_t total_tiles,
+                      uint32_t tiles_per_part_unused, uint32_t nstripes,
                       hipStream_t s) {
+	constexpr int CH = ec_chunks_for(D);
+	uint32_t tile_bytes = kChunkBytes * CH;
+	uint32_t tiles_per_part = (part_len + tile_bytes - 1) / tile_bytes;
+	uint32_t total_tiles = tiles_per_part * nstripes;
 	uint32_t grid = total_tiles < 262144u ? total_tiles : 262144u;
 	size_t lds = (size_t)D * srcs * 32;
-	hipLaunchKernelGGL(HIP_KERNEL_NAME(ec_encode_kernel<D, kECChunks, kECSwz, kECNtStore>), dim3(grid), dim3(kThreads), lds, s,
+	(void)tiles_per_part_unused;
+	hipLaunchKernelGGL(HIP_KERNEL_NAME(ec_encode_kernel<D, CH, kECSwz, kECNtStore>),
+	                   dim3(grid), dim3(kThreads), lds, s,
 	                   part_len, srcs, dest_base, d_tbls, d_src, d_dst,
 	                   dests_total, tiles_per_part, total_tiles);
 }
 
 static int run_batch(uint64_t part_len, int srcs, int dests,
                      const uint8_t *d_tbls, const uint64_t *d_src,
-                     const uint64_t *d_dst, uint32_t tiles_per_part,
-                     uint32_t total_tiles, hipStream_t s) {
+                     const uint64_t *d_dst, uint32_t num_stripes,
+                     hipStream_t s) {
 	/* One pass computes up to kMaxDestsPerPass destinations; wider m splits
 	 * into passes (each pass re-reads the sources, so fewer passes wins on
 	 * this HBM-bound kernel — D<=8 measured best, profiles/). */
@@ -368,28 +374,28 @@ static int run_batch(uint64_t part_len, int srcs, int dests,
 		if (d > 8) d = 8;
 		switch (d) {
 		case 1: launch_ec<1>((uint32_t)part_len, srcs, base, d_tbls, d_src,
-		                     d_dst, dests, tiles_per_part, total_tiles, s);
+		                     d_dst, dests, 0, num_stripes, s);
 			break;
 		case 2: launch_ec<2>((uint32_t)part_len, srcs, base, d_tbls, d_src,
-		                     d_dst, dests, tiles_per_part, total_tiles, s);
+		                     d_dst, dests, 0, num_stripes, s);
 			break;
 		case 3: launch_ec<3>((uint32_t)part_len, srcs, base, d_tbls, d_src,
-		                     d_dst, dests, tiles_per_part, total_tiles, s);
+		                     d_dst, dests, 0, num_stripes, s);
 			break;
 		case 4: launch_ec<4>((uint32_t)part_len, srcs, base, d_tbls, d_src,
-		                     d_dst, dests, tiles_per_part, total_tiles, s);
+		                     d_dst, dests, 0, num_stripes, s);
 			break;
 		case 5: launch_ec<5>((uint32_t)part_len, srcs, base, d_tbls, d_src,
-		                     d_dst, dests, tiles_per_part, total_tiles, s);
+		                     d_dst, dests, 0, num_stripes, s);
 			break;
 		case 6: launch_ec<6>((uint32_t)part_len, srcs, base, d_tbls, d_src,
-		                     d_dst, dests, tiles_per_part, total_tiles, s);
+		                     d_dst, dests, 0, num_stripes, s);
 			break;
 		case 7: launch_ec<7>((uint32_t)part_len, srcs, base, d_tbls, d_src,
-		                     d_dst, dests, tiles_per_part, total_tiles, s);
+		                     d_dst, dests, 0, num_stripes, s);
 			break;
 		default: launch_ec<8>((uint32_t)part_len, srcs, base, d_tbls, d_src,
-		                      d_dst, dests, tiles_per_part, total_tiles, s);
+		                      d_dst, dests, 0, num_stripes, s);
 			break;
 		}
 		base += d;
@@ -423,11 +429,8 @@ extern "C" int lizec_ec_encode_batch(lizec_engine *e, uint64_t part_len,
 	LIZEC_CHECK(hipMemcpyAsync(d_dst, dst_dptrs, ndst * 8,
 	                           hipMemcpyHostToDevice, s));
 
-	uint32_t tiles_per_part =
-	    (uint32_t)((part_len + kTileBytes - 1) / kTileBytes);
-	uint32_t total_tiles = tiles_per_part * (uint32_t)num_stripes;
 	return run_batch(part_len, srcs, dests, e->d_gftbls, d_src, d_dst,
-	                 tiles_per_part, total_tiles, s);
+	                 (uint32_t)num_stripes, s);
 }
 
 extern "C" int lizec_ec_plan_create(lizec_engine *e, uint64_t part_len,
@@ -447,8 +450,6 @@ extern "C" int lizec_ec_plan_create(lizec_engine *e, uint64_t part_len,
 	p->srcs = srcs;
 	p->dests = dests;
 	p->num_stripes = num_stripes;
-	p->tiles_per_part = (uint32_t)((part_len + kTileBytes - 1) / kTileBytes);
-	p->total_tiles = p->tiles_per_part * (uint32_t)num_stripes;
 	size_t nsrc = (size_t)num_stripes * srcs;
 	size_t ndst = (size_t)num_stripes * dests;
 	if (hipMalloc(&p->d_tbls, (size_t)32 * srcs * dests) != hipSuccess ||
@@ -475,7 +476,7 @@ extern "C" int lizec_ec_plan_run(lizec_plan *p, void *stream) {
 	hipStream_t s = stream ? (hipStream_t)stream : p->e->stream;
 	LIZEC_CHECK(hipSetDevice(p->e->device));
 	return run_batch(p->part_len, p->srcs, p->dests, p->d_tbls, p->d_src,
-	                 p->d_dst, p->tiles_per_part, p->total_tiles, s);
+	                 p->d_dst, (uint32_t)p->num_stripes, s);
 }
 
 extern "C" void lizec_ec_plan_destroy(lizec_plan *p) {
