@@ -135,14 +135,17 @@ __global__ __launch_bounds__(kThreads) void crc32_blocks_kernel(
 }
 
 /* Fast path (block_len % 16384 == 0, e.g. the standard 64 KiB block):
- * one wave per block, TWO independent chains per lane (lane l owns segment
- * l of the first half and segment l of the second half, 128 segments of
- * block_len/128 bytes) — doubles the serial-dependency ILP of the CRC
- * recurrence — with full 128-byte line bursts per chain so every fetched
- * line is consumed while resident (fixes the 2.2x HBM over-fetch the PMC
- * counters showed for the strided 16-byte walk), and slicing-by-8 tables
- * (half the dependent LDS steps of slicing-by-4). */
-__global__ __launch_bounds__(kThreads) void crc32_blocks_kernel_fast(
+ * one wave per block, C independent CRC chains per lane.  The block is cut
+ * into C equal spans; lane l owns segment l of every span, so each lane
+ * advances C serial CRC recurrences at once (ILP) and bursts BV*16
+ * contiguous bytes per chain per iteration so every fetched line is
+ * consumed while resident (the naive strided walk over-fetched HBM 2.2x
+ * per the PMC counters).  Slicing-by-8 tables halve the dependent LDS
+ * steps; per-lane CRCs fold in a shfl tree using the GF(2) "advance by N
+ * zero bytes" matrices of mycrc32_combine (crc.cc:153-224), then lane 0
+ * splices the C span CRCs. */
+template <int C, int BV>
+__global__ __launch_bounds__(kThreads) void crc32_blocks_kernel_multi(
     const uint8_t *__restrict__ buf, uint32_t block_len, uint64_t nblocks,
     uint32_t seed, const uint32_t *__restrict__ crc_const,
     uint32_t *__restrict__ out) {
@@ -155,22 +158,23 @@ __global__ __launch_bounds__(kThreads) void crc32_blocks_kernel_fast(
 
 	const int wave = threadIdx.x >> 6;
 	const int lane = threadIdx.x & 63;
-	const uint32_t seg = block_len >> 7;   /* bytes per chain segment */
-	const uint32_t half = block_len >> 1;
+	const uint32_t span = block_len / C;
+	const uint32_t seg = span >> 6;        /* bytes per chain segment */
 
 	for (uint64_t blk = (uint64_t)blockIdx.x * 4 + wave; blk < nblocks;
 	     blk += (uint64_t)gridDim.x * 4) {
-		const uint8_t *pa = buf + blk * block_len + (uint32_t)lane * seg;
-		const uint8_t *pb = pa + half;
-		uint32_t ca = (lane == 0 ? seed : 0u) ^ 0xFFFFFFFFu;
-		uint32_t cb = 0xFFFFFFFFu;
-		for (uint32_t i = 0; i < seg; i += 128) {
-			uint4 wa[8], wb[8];
+		const uint8_t *base = buf + blk * block_len + (uint32_t)lane * seg;
+		uint32_t crc[C];
 #pragma unroll
-			for (int q = 0; q < 8; ++q) {
-				wa[q] = *(const uint4 *)(pa + i + q * 16);
-				wb[q] = *(const uint4 *)(pb + i + q * 16);
-			}
+		for (int c = 0; c < C; ++c)
+			crc[c] = ((c == 0 && lane == 0) ? seed : 0u) ^ 0xFFFFFFFFu;
+		for (uint32_t i = 0; i < seg; i += 16 * BV) {
+			uint4 w[C][BV];
+#pragma unroll
+			for (int c = 0; c < C; ++c)
+#pragma unroll
+				for (int q = 0; q < BV; ++q)
+					w[c][q] = *(const uint4 *)(base + c * span + i + q * 16);
 #define LIZEC_CRC8(crc, lo, hi)                                          \
 	do {                                                                 \
 		uint32_t u0 = (crc) ^ (lo), u1 = (hi);                           \
@@ -183,29 +187,36 @@ __global__ __launch_bounds__(kThreads) void crc32_blocks_kernel_fast(
 		        T[1 * 256 + ((u1 >> 16) & 0xff)] ^ T[u1 >> 24];          \
 	} while (0)
 #pragma unroll
-			for (int q = 0; q < 8; ++q) {
-				LIZEC_CRC8(ca, wa[q].x, wa[q].y);
-				LIZEC_CRC8(cb, wb[q].x, wb[q].y);
-				LIZEC_CRC8(ca, wa[q].z, wa[q].w);
-				LIZEC_CRC8(cb, wb[q].z, wb[q].w);
-			}
+			for (int q = 0; q < BV; ++q)
+#pragma unroll
+				for (int c = 0; c < C; ++c) {
+					LIZEC_CRC8(crc[c], w[c][q].x, w[c][q].y);
+					LIZEC_CRC8(crc[c], w[c][q].z, w[c][q].w);
+				}
 #undef LIZEC_CRC8
 		}
-		ca ^= 0xFFFFFFFFu;
-		cb ^= 0xFFFFFFFFu;
+#pragma unroll
+		for (int c = 0; c < C; ++c) crc[c] ^= 0xFFFFFFFFu;
 
-		/* fold each half's 64 segments, then splice the halves */
+		/* fold each span's 64 segments, then splice the spans */
 		uint32_t len = seg;
 #pragma unroll
 		for (int s = 0; s < 6; ++s) {
-			uint32_t oa = __shfl_down(ca, 1 << s, 64);
-			uint32_t ob = __shfl_down(cb, 1 << s, 64);
 			uint32_t olen = __shfl_down(len, 1 << s, 64);
-			ca = crc_advance(ca, olen, mats) ^ oa;
-			cb = crc_advance(cb, olen, mats) ^ ob;
+#pragma unroll
+			for (int c = 0; c < C; ++c) {
+				uint32_t o = __shfl_down(crc[c], 1 << s, 64);
+				crc[c] = crc_advance(crc[c], olen, mats) ^ o;
+			}
 			len += olen;
 		}
-		if (lane == 0) out[blk] = crc_advance(ca, half, mats) ^ cb;
+		if (lane == 0) {
+			uint32_t acc = crc[0];
+#pragma unroll
+			for (int c = 1; c < C; ++c)
+				acc = crc_advance(acc, span, mats) ^ crc[c];
+			out[blk] = acc;
+		}
 	}
 }
 
@@ -498,11 +509,18 @@ extern "C" int lizec_crc32_batch(lizec_engine *e, const void *dev_buf,
 	LIZEC_CHECK(hipSetDevice(e->device));
 	uint64_t groups = (nblocks + 3) / 4;
 	uint32_t grid = (uint32_t)(groups < 131072 ? groups : 131072);
-	if (block_len % 16384 == 0)
-		hipLaunchKernelGGL(crc32_blocks_kernel_fast, dim3(grid),
-		                   dim3(kThreads), 0, s, (const uint8_t *)dev_buf,
-		                   block_len, nblocks, seed, e->d_crc_const,
-		                   dev_crcs_out);
+	const char *ch = getenv("LIZEC_CRC_CHAINS");   /* A/B hook */
+	int chains = ch ? atoi(ch) : 4;
+	if (block_len % 16384 == 0 && chains >= 4)
+		hipLaunchKernelGGL(HIP_KERNEL_NAME(crc32_blocks_kernel_multi<4, 4>),
+		                   dim3(grid), dim3(kThreads), 0, s,
+		                   (const uint8_t *)dev_buf, block_len, nblocks,
+		                   seed, e->d_crc_const, dev_crcs_out);
+	else if (block_len % 16384 == 0)
+		hipLaunchKernelGGL(HIP_KERNEL_NAME(crc32_blocks_kernel_multi<2, 8>),
+		                   dim3(grid), dim3(kThreads), 0, s,
+		                   (const uint8_t *)dev_buf, block_len, nblocks,
+		                   seed, e->d_crc_const, dev_crcs_out);
 	else
 		hipLaunchKernelGGL(crc32_blocks_kernel, dim3(grid), dim3(kThreads),
 		                   0, s, (const uint8_t *)dev_buf, block_len, nblocks,
