@@ -510,7 +510,7 @@ extern "C" int lizec_crc32_batch(lizec_engine *e, const void *dev_buf,
 	uint64_t groups = (nblocks + 3) / 4;
 	uint32_t grid = (uint32_t)(groups < 131072 ? groups : 131072);
 	const char *ch = getenv("LIZEC_CRC_CHAINS");   /* A/B hook */
-	int chains = ch ? atoi(ch) : 4;
+	int chains = ch ? atoi(ch) : 2;   /* C=4 measured worse: half-line bursts re-thrash L1 (profiles) */
 	if (block_len % 16384 == 0 && chains >= 4)
 		hipLaunchKernelGGL(HIP_KERNEL_NAME(crc32_blocks_kernel_multi<4, 4>),
 		                   dim3(grid), dim3(kThreads), 0, s,
