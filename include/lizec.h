@@ -183,6 +183,21 @@ int lizec_crc32_batch(lizec_engine *e, const void *dev_buf, uint32_t block_len,
                       uint64_t nblocks, uint32_t seed, uint32_t *dev_crcs_out,
                       void *stream);
 
+/* Batched chunk scrub — hdd_int_test semantics (hddspacemgr.cc:2148-2212)
+ * over MooseFS-format chunk-part images (chunk.cc:126-188: 1 KiB signature
+ * block, big-endian u32 CRC array at crc_off, 64 KiB blocks at data_off):
+ * verifies every block's CRC against the stored array.
+ *  chunk_dptrs[i] : device address of part-file image i
+ *  data_offs[i]   : header size (offset of block 0)
+ *  crc_offs[i]    : offset of the CRC array (kMaxSignatureBlockSize = 1024)
+ *  block_counts[i]: blocks in image i
+ *  dev_status_out : device int32[nchunks]; first damaged block index, or
+ *                   INT32_MAX if the image is clean. */
+int lizec_scrub_batch(lizec_engine *e, const uint64_t *chunk_dptrs,
+                      const uint32_t *data_offs, const uint32_t *crc_offs,
+                      const uint32_t *block_counts, int nchunks,
+                      int32_t *dev_status_out, void *stream);
+
 #ifdef __cplusplus
 }
 #endif

@@ -144,37 +144,28 @@ __global__ __launch_bounds__(kThreads) void crc32_blocks_kernel(
  * steps; per-lane CRCs fold in a shfl tree using the GF(2) "advance by N
  * zero bytes" matrices of mycrc32_combine (crc.cc:153-224), then lane 0
  * splices the C span CRCs. */
+/* Per-wave CRC of one block: C chains x 64 lane segments, BV*16-byte
+ * bursts, slicing-by-8; returns the block CRC on every lane (lane 0's
+ * value is authoritative). */
 template <int C, int BV>
-__global__ __launch_bounds__(kThreads) void crc32_blocks_kernel_multi(
-    const uint8_t *__restrict__ buf, uint32_t block_len, uint64_t nblocks,
-    uint32_t seed, const uint32_t *__restrict__ crc_const,
-    uint32_t *__restrict__ out) {
-	__shared__ __attribute__((aligned(16))) uint32_t stabs[kCrcConstWords];
-	for (int i = threadIdx.x; i < kCrcConstWords; i += kThreads)
-		stabs[i] = crc_const[i];
-	__syncthreads();
-	const uint32_t *T = stabs;            /* T[t*256 + i], t = 0..7 */
-	const uint32_t *mats = stabs + kCrcTabWords;
-
-	const int wave = threadIdx.x >> 6;
-	const int lane = threadIdx.x & 63;
+__device__ uint32_t crc_block_wave(const uint8_t *__restrict__ block,
+                                   uint32_t block_len, uint32_t seed,
+                                   const uint32_t *T, const uint32_t *mats,
+                                   int lane) {
 	const uint32_t span = block_len / C;
-	const uint32_t seg = span >> 6;        /* bytes per chain segment */
-
-	for (uint64_t blk = (uint64_t)blockIdx.x * 4 + wave; blk < nblocks;
-	     blk += (uint64_t)gridDim.x * 4) {
-		const uint8_t *base = buf + blk * block_len + (uint32_t)lane * seg;
-		uint32_t crc[C];
+	const uint32_t seg = span >> 6;
+	const uint8_t *base = block + (uint32_t)lane * seg;
+	uint32_t crc[C];
+#pragma unroll
+	for (int c = 0; c < C; ++c)
+		crc[c] = ((c == 0 && lane == 0) ? seed : 0u) ^ 0xFFFFFFFFu;
+	for (uint32_t i = 0; i < seg; i += 16 * BV) {
+		uint4 w[C][BV];
 #pragma unroll
 		for (int c = 0; c < C; ++c)
-			crc[c] = ((c == 0 && lane == 0) ? seed : 0u) ^ 0xFFFFFFFFu;
-		for (uint32_t i = 0; i < seg; i += 16 * BV) {
-			uint4 w[C][BV];
 #pragma unroll
-			for (int c = 0; c < C; ++c)
-#pragma unroll
-				for (int q = 0; q < BV; ++q)
-					w[c][q] = *(const uint4 *)(base + c * span + i + q * 16);
+			for (int q = 0; q < BV; ++q)
+				w[c][q] = *(const uint4 *)(base + c * span + i + q * 16);
 #define LIZEC_CRC8(crc, lo, hi)                                          \
 	do {                                                                 \
 		uint32_t u0 = (crc) ^ (lo), u1 = (hi);                           \
@@ -187,35 +178,92 @@ __global__ __launch_bounds__(kThreads) void crc32_blocks_kernel_multi(
 		        T[1 * 256 + ((u1 >> 16) & 0xff)] ^ T[u1 >> 24];          \
 	} while (0)
 #pragma unroll
-			for (int q = 0; q < BV; ++q)
-#pragma unroll
-				for (int c = 0; c < C; ++c) {
-					LIZEC_CRC8(crc[c], w[c][q].x, w[c][q].y);
-					LIZEC_CRC8(crc[c], w[c][q].z, w[c][q].w);
-				}
-#undef LIZEC_CRC8
-		}
-#pragma unroll
-		for (int c = 0; c < C; ++c) crc[c] ^= 0xFFFFFFFFu;
-
-		/* fold each span's 64 segments, then splice the spans */
-		uint32_t len = seg;
-#pragma unroll
-		for (int s = 0; s < 6; ++s) {
-			uint32_t olen = __shfl_down(len, 1 << s, 64);
+		for (int q = 0; q < BV; ++q)
 #pragma unroll
 			for (int c = 0; c < C; ++c) {
-				uint32_t o = __shfl_down(crc[c], 1 << s, 64);
-				crc[c] = crc_advance(crc[c], olen, mats) ^ o;
+				LIZEC_CRC8(crc[c], w[c][q].x, w[c][q].y);
+				LIZEC_CRC8(crc[c], w[c][q].z, w[c][q].w);
 			}
-			len += olen;
-		}
-		if (lane == 0) {
-			uint32_t acc = crc[0];
+#undef LIZEC_CRC8
+	}
 #pragma unroll
-			for (int c = 1; c < C; ++c)
-				acc = crc_advance(acc, span, mats) ^ crc[c];
-			out[blk] = acc;
+	for (int c = 0; c < C; ++c) crc[c] ^= 0xFFFFFFFFu;
+
+	uint32_t len = seg;
+#pragma unroll
+	for (int s = 0; s < 6; ++s) {
+		uint32_t olen = __shfl_down(len, 1 << s, 64);
+#pragma unroll
+		for (int c = 0; c < C; ++c) {
+			uint32_t o = __shfl_down(crc[c], 1 << s, 64);
+			crc[c] = crc_advance(crc[c], olen, mats) ^ o;
+		}
+		len += olen;
+	}
+	uint32_t acc = crc[0];
+#pragma unroll
+	for (int c = 1; c < C; ++c)
+		acc = crc_advance(acc, span, mats) ^ crc[c];
+	return acc;
+}
+
+template <int C, int BV>
+__global__ __launch_bounds__(kThreads) void crc32_blocks_kernel_multi(
+    const uint8_t *__restrict__ buf, uint32_t block_len, uint64_t nblocks,
+    uint32_t seed, const uint32_t *__restrict__ crc_const,
+    uint32_t *__restrict__ out) {
+	__shared__ __attribute__((aligned(16))) uint32_t stabs[kCrcConstWords];
+	for (int i = threadIdx.x; i < kCrcConstWords; i += kThreads)
+		stabs[i] = crc_const[i];
+	__syncthreads();
+	const uint32_t *T = stabs;
+	const uint32_t *mats = stabs + kCrcTabWords;
+	const int wave = threadIdx.x >> 6;
+	const int lane = threadIdx.x & 63;
+	for (uint64_t blk = (uint64_t)blockIdx.x * 4 + wave; blk < nblocks;
+	     blk += (uint64_t)gridDim.x * 4) {
+		uint32_t crc = crc_block_wave<C, BV>(buf + blk * block_len, block_len,
+		                                     seed, T, mats, lane);
+		if (lane == 0) out[blk] = crc;
+	}
+}
+
+
+/* Batched chunk scrub — hdd_int_test semantics (hddspacemgr.cc:2148-2212):
+ * for every 64 KiB block of every MooseFS-format chunk-part image, compare
+ * mycrc32(0, block, MFSBLOCKSIZE) against the stored CRC array entry
+ * (big-endian u32 at crc_off + 4*b, cf. get32bit at hddspacemgr.cc:2183 and
+ * chunk.cc:183-188).  status[c] collects the FIRST damaged block index via
+ * atomicMin (host pre-fills INT32_MAX = clean). */
+__global__ __launch_bounds__(kThreads) void scrub_chunks_kernel(
+    const uint64_t *__restrict__ chunk_dptrs,
+    const uint32_t *__restrict__ data_offs,
+    const uint32_t *__restrict__ crc_offs,
+    const uint32_t *__restrict__ block_counts, uint32_t nchunks,
+    uint32_t max_blocks, const uint32_t *__restrict__ crc_const,
+    int32_t *__restrict__ status) {
+	__shared__ __attribute__((aligned(16))) uint32_t stabs[kCrcConstWords];
+	for (int i = threadIdx.x; i < kCrcConstWords; i += kThreads)
+		stabs[i] = crc_const[i];
+	__syncthreads();
+	const uint32_t *T = stabs;
+	const uint32_t *mats = stabs + kCrcTabWords;
+	const int wave = threadIdx.x >> 6;
+	const int lane = threadIdx.x & 63;
+	const uint64_t total = (uint64_t)nchunks * max_blocks;
+	for (uint64_t flat = (uint64_t)blockIdx.x * 4 + wave; flat < total;
+	     flat += (uint64_t)gridDim.x * 4) {
+		uint32_t c = (uint32_t)(flat / max_blocks);
+		uint32_t b = (uint32_t)(flat - (uint64_t)c * max_blocks);
+		if (b >= block_counts[c]) continue;
+		const uint8_t *img = (const uint8_t *)chunk_dptrs[c];
+		uint32_t crc = crc_block_wave<2, 8>(img + data_offs[c] + b * 65536u,
+		                                    65536u, 0u, T, mats, lane);
+		if (lane == 0) {
+			const uint8_t *p = img + crc_offs[c] + 4u * b;
+			uint32_t stored = ((uint32_t)p[0] << 24) | ((uint32_t)p[1] << 16) |
+			                  ((uint32_t)p[2] << 8) | p[3];
+			if (stored != crc) atomicMin(&status[c], (int32_t)b);
 		}
 	}
 }
@@ -525,6 +573,49 @@ extern "C" int lizec_crc32_batch(lizec_engine *e, const void *dev_buf,
 		hipLaunchKernelGGL(crc32_blocks_kernel, dim3(grid), dim3(kThreads),
 		                   0, s, (const uint8_t *)dev_buf, block_len, nblocks,
 		                   seed, e->d_crc_const, dev_crcs_out);
+	LIZEC_CHECK(hipGetLastError());
+	return LIZEC_OK;
+}
+
+extern "C" int lizec_scrub_batch(lizec_engine *e, const uint64_t *chunk_dptrs,
+                                 const uint32_t *data_offs,
+                                 const uint32_t *crc_offs,
+                                 const uint32_t *block_counts, int nchunks,
+                                 int32_t *dev_status_out, void *stream) {
+	if (!e || !chunk_dptrs || nchunks < 1 || !dev_status_out)
+		return LIZEC_EINVAL;
+	hipStream_t s = stream ? (hipStream_t)stream : e->stream;
+	LIZEC_CHECK(hipSetDevice(e->device));
+	/* upload per-chunk tables into engine scratch (u64 slots reused) */
+	size_t words = (size_t)nchunks;               /* dptrs */
+	size_t meta = (words * 8 + 3 * words * 4 + 7) / 8 + words;
+	int r = ensure_ptrs(e, meta + 8);
+	if (r != LIZEC_OK) return r;
+	uint64_t *d_ptrs = e->d_ptrs;
+	uint32_t *d_doffs = (uint32_t *)(d_ptrs + nchunks);
+	uint32_t *d_coffs = d_doffs + nchunks;
+	uint32_t *d_counts = d_coffs + nchunks;
+	LIZEC_CHECK(hipMemcpyAsync(d_ptrs, chunk_dptrs, (size_t)nchunks * 8,
+	                           hipMemcpyHostToDevice, s));
+	LIZEC_CHECK(hipMemcpyAsync(d_doffs, data_offs, (size_t)nchunks * 4,
+	                           hipMemcpyHostToDevice, s));
+	LIZEC_CHECK(hipMemcpyAsync(d_coffs, crc_offs, (size_t)nchunks * 4,
+	                           hipMemcpyHostToDevice, s));
+	LIZEC_CHECK(hipMemcpyAsync(d_counts, block_counts, (size_t)nchunks * 4,
+	                           hipMemcpyHostToDevice, s));
+	uint32_t max_blocks = 0;
+	for (int i = 0; i < nchunks; ++i)
+		if (block_counts[i] > max_blocks) max_blocks = block_counts[i];
+	if (max_blocks == 0) return LIZEC_EINVAL;
+	/* INT32_MAX sentinel = clean */
+	LIZEC_CHECK(hipMemsetD32Async((hipDeviceptr_t)dev_status_out, 0x7FFFFFFF,
+	                              nchunks, s));
+	uint64_t total = (uint64_t)nchunks * max_blocks;
+	uint64_t groups = (total + 3) / 4;
+	uint32_t grid = (uint32_t)(groups < 131072 ? groups : 131072);
+	hipLaunchKernelGGL(scrub_chunks_kernel, dim3(grid), dim3(kThreads), 0, s,
+	                   d_ptrs, d_doffs, d_coffs, d_counts, (uint32_t)nchunks,
+	                   max_blocks, e->d_crc_const, dev_status_out);
 	LIZEC_CHECK(hipGetLastError());
 	return LIZEC_OK;
 }

@@ -85,6 +85,12 @@ def lib():
     L.lizec_crc32_batch.argtypes = [
         ctypes.c_void_p, ctypes.c_void_p, ctypes.c_uint32, ctypes.c_uint64,
         ctypes.c_uint32, ctypes.c_void_p, ctypes.c_void_p]
+    L.lizec_scrub_batch.restype = ctypes.c_int
+    L.lizec_scrub_batch.argtypes = [
+        ctypes.c_void_p, ctypes.POINTER(ctypes.c_uint64),
+        ctypes.POINTER(ctypes.c_uint32), ctypes.POINTER(ctypes.c_uint32),
+        ctypes.POINTER(ctypes.c_uint32), ctypes.c_int, ctypes.c_void_p,
+        ctypes.c_void_p]
     _lib = L
     return _lib
 
