@@ -75,6 +75,19 @@ uint32_t lizec_crc32(uint32_t crc, const uint8_t *block, uint32_t leng);
 uint32_t lizec_crc32_combine(uint32_t crc1, uint32_t crc2, uint32_t leng2);
 void lizec_crc32_init(void);
 
+/* Partial-block CRC algebra (crc.h:27-29 macros + hdd_write's splice,
+ * hddspacemgr.cc:1952-2003) — byte-range writes without re-hashing. */
+uint32_t lizec_crc32_zeroblock(uint32_t crc, uint32_t zeros);
+uint32_t lizec_crc32_zeroexpanded(uint32_t crc, const uint8_t *block,
+                                  uint32_t leng, uint32_t zeros);
+uint32_t lizec_crc32_xorblocks(uint32_t crc, uint32_t crcblock1,
+                               uint32_t crcblock2, uint32_t leng);
+uint32_t lizec_crc32_splice(uint32_t precrc, uint32_t offset, uint32_t crc,
+                            uint32_t size, uint32_t postcrc,
+                            uint32_t block_len);
+void lizec_recompute_crc_if_block_empty(const uint8_t *block,
+                                        uint32_t block_len, uint32_t *crc);
+
 /* Host byte-XOR (xor-goal parity building block; also exported with the
  * C++ mangling of the reference's blockXor, common/block_xor.h:33).
  * GPU xor parity/recovery runs through the EC kernel with all-ones

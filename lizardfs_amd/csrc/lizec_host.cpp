@@ -325,6 +325,54 @@ extern "C" uint32_t lizec_crc32_combine(uint32_t crc1, uint32_t crc2,
 	return crc1 ^ crc2;
 }
 
+/* Partial-block CRC algebra — the crc.h:27-29 macros and hdd_write's
+ * splice logic (hddspacemgr.cc:1952-2003) as first-class functions. */
+
+/* crc.h:27: mycrc32_zeroblock — CRC after appending `zeros` zero bytes. */
+extern "C" uint32_t lizec_crc32_zeroblock(uint32_t crc, uint32_t zeros) {
+	return lizec_crc32_combine(crc ^ 0xFFFFFFFFu, 0xFFFFFFFFu, zeros);
+}
+
+/* crc.h:28: mycrc32_zeroexpanded — CRC of data followed by zeros. */
+extern "C" uint32_t lizec_crc32_zeroexpanded(uint32_t crc, const uint8_t *block,
+                                             uint32_t leng, uint32_t zeros) {
+	return lizec_crc32_zeroblock(lizec_crc32(crc, block, leng), zeros);
+}
+
+/* crc.h:29: mycrc32_xorblocks — CRC of the byte-XOR of two equal-length
+ * blocks from their CRCs (CRC is affine over GF(2)). */
+extern "C" uint32_t lizec_crc32_xorblocks(uint32_t crc, uint32_t crcblock1,
+                                          uint32_t crcblock2, uint32_t leng) {
+	return crcblock1 ^ crcblock2 ^ lizec_crc32_zeroblock(crc, leng);
+}
+
+/* hdd_write's partial-write recombine (hddspacemgr.cc:1995-2003): CRC of a
+ * block_len-byte block whose [offset, offset+size) range has CRC `crc`,
+ * with precrc = CRC of [0, offset) and postcrc = CRC of the tail. */
+extern "C" uint32_t lizec_crc32_splice(uint32_t precrc, uint32_t offset,
+                                       uint32_t crc, uint32_t size,
+                                       uint32_t postcrc, uint32_t block_len) {
+	uint32_t combined;
+	if (offset == 0) {
+		return lizec_crc32_combine(crc, postcrc, block_len - size);
+	}
+	combined = lizec_crc32_combine(precrc, crc, size);
+	if (offset + size < block_len)
+		combined = lizec_crc32_combine(combined, postcrc,
+		                               block_len - (offset + size));
+	return combined;
+}
+
+/* crc.cc:235-243: sparse-file special case — a zero block whose cached
+ * CRC is 0 gets the canonical empty-block CRC. */
+extern "C" void lizec_recompute_crc_if_block_empty(const uint8_t *block,
+                                                   uint32_t block_len,
+                                                   uint32_t *crc) {
+	if (*crc != 0) return;
+	if (block[0] != 0 || memcmp(block, block + 1, block_len - 1) != 0) return;
+	*crc = lizec_crc32_zeroblock(0, block_len);
+}
+
 /* C++-linkage aliases so a LizardFS build linking against common/crc.h's
  * mangled symbols (crc.h:25-31) resolves them from this library. */
 uint32_t mycrc32(uint32_t crc, const uint8_t *block, uint32_t leng) {
