@@ -76,6 +76,9 @@ class ReedSolomon:
         if plan_key is not None:
             plan = self._plans.get(plan_key)
             if plan is None:
+                while len(self._plans) >= 64:   # bound device-side state
+                    oldest = next(iter(self._plans))
+                    self._lib.lizec_ec_plan_destroy(self._plans.pop(oldest))
                 plan = ctypes.c_void_p()
                 L.check(self._lib.lizec_ec_plan_create(
                     self._engine, part_len, ic, oc,
