@@ -47,7 +47,7 @@ struct Variant {
 	int ch;
 };
 
-template <int D, int CH, bool SWZ, bool NTST>
+template <int D, int CH, bool SWZ, bool NTST, bool NTLD = false>
 static void launch_var(uint32_t part_len, int srcs, const uint8_t *tbls,
                        const uint64_t *src, const uint64_t *dst, int dests,
                        uint32_t tiles_per_part, uint32_t total_tiles,
@@ -55,7 +55,7 @@ static void launch_var(uint32_t part_len, int srcs, const uint8_t *tbls,
 	uint32_t grid = total_tiles < grid_cap ? total_tiles : grid_cap;
 	size_t lds = (size_t)D * srcs * 32;
 	for (int base = 0; base + D <= dests; base += D)
-		hipLaunchKernelGGL(HIP_KERNEL_NAME(ec_encode_kernel<D, CH, SWZ, NTST>),
+		hipLaunchKernelGGL(HIP_KERNEL_NAME(ec_encode_kernel<D, CH, SWZ, NTST, NTLD>),
 		                   dim3(grid), dim3(kThreads), lds, s, part_len, srcs,
 		                   base, tbls, src, dst, dests, tiles_per_part,
 		                   total_tiles);
@@ -125,6 +125,8 @@ int main(int argc, char **argv) {
 	Cfg cfgs[] = {
 	    {"D1_CH4_swz_nt ", launch_var<1, 4, true, true>, 4, 1, 262144},
 	    {"D2_CH4_swz_nt ", launch_var<2, 4, true, true>, 4, 2, 262144},
+	    {"D2_CH4_swz_nt_ntld", launch_var<2, 4, true, true, true>, 4, 2, 262144},
+	    {"D2_CH6_swz_nt ", launch_var<2, 6, true, true>, 6, 2, 262144},
 	    {"D2_CH4_base   ", launch_var<2, 4, false, false>, 4, 2, 262144},
 	    {"D3_CH4_swz_nt ", launch_var<3, 4, true, true>, 4, 3, 262144},
 	    {"D4_CH4_swz_nt ", launch_var<4, 4, true, true>, 4, 4, 262144},

@@ -65,7 +65,14 @@ __device__ __forceinline__ uint32_t xcd_remap(uint32_t b, uint32_t n) {
 	return (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + i;
 }
 
-template <int D, int CH, bool SWZ, bool NTST>
+/* non-temporal 16B load helper (streamed sources, no reuse) */
+__device__ __forceinline__ uint4 ld_nt(const uint8_t *p) {
+	typedef unsigned int u32x4 __attribute__((ext_vector_type(4)));
+	u32x4 v = __builtin_nontemporal_load((const u32x4 *)p);
+	return make_uint4(v.x, v.y, v.z, v.w);
+}
+
+template <int D, int CH, bool SWZ, bool NTST, bool NTLD = false>
 __global__ __launch_bounds__(kThreads) void ec_encode_kernel(
     uint32_t part_len, int srcs, int dest_base,
     const uint8_t *__restrict__ gftbls_dev,  /* 32*srcs*dests_total */
@@ -109,14 +116,16 @@ __global__ __launch_bounds__(kThreads) void ec_encode_kernel(
 				const uint8_t *sp = (const uint8_t *)sp_tab[0];
 #pragma unroll
 				for (int c = 0; c < CH; ++c)
-					w[c] = *(const uint4 *)(sp + (base + c * kChunkBytes));
+					w[c] = NTLD ? ld_nt(sp + (base + c * kChunkBytes))
+					            : *(const uint4 *)(sp + (base + c * kChunkBytes));
 			}
 			for (int j = 0; j < srcs; ++j) {
 				if (j + 1 < srcs) {
 					const uint8_t *spn = (const uint8_t *)sp_tab[j + 1];
 #pragma unroll
 					for (int c = 0; c < CH; ++c)
-						wn[c] = *(const uint4 *)(spn + (base + c * kChunkBytes));
+						wn[c] = NTLD ? ld_nt(spn + (base + c * kChunkBytes))
+						             : *(const uint4 *)(spn + (base + c * kChunkBytes));
 				}
 				uint4 L[D], H[D];
 #pragma unroll
