@@ -560,8 +560,8 @@ extern "C" int lizec_crc32_batch(lizec_engine *e, const void *dev_buf,
 	uint32_t grid = (uint32_t)(groups < 131072 ? groups : 131072);
 	const char *ch = getenv("LIZEC_CRC_CHAINS");   /* A/B hook */
 	int chains = ch ? atoi(ch) : 2;   /* C=4 measured worse: half-line bursts re-thrash L1 (profiles) */
-	if (block_len % 16384 == 0 && chains >= 4)
-		hipLaunchKernelGGL(HIP_KERNEL_NAME(crc32_blocks_kernel_multi<4, 4>),
+	if (block_len % 32768 == 0 && chains >= 4)
+		hipLaunchKernelGGL(HIP_KERNEL_NAME(crc32_blocks_kernel_multi<4, 8>),
 		                   dim3(grid), dim3(kThreads), 0, s,
 		                   (const uint8_t *)dev_buf, block_len, nblocks,
 		                   seed, e->d_crc_const, dev_crcs_out);
