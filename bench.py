@@ -142,6 +142,11 @@ def main():
     world = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    if args.gpus > 1 and world == 1:
+        log(f"--gpus {args.gpus} requires torchrun (one rank per GPU): "
+            f"python -m torch.distributed.run --nnodes=1 --nproc-per-node "
+            f"{args.gpus} --master-addr 127.0.0.1 bench.py ...")
+        sys.exit(2)
     n_gpus = world if world > 1 else args.gpus
     distributed = world > 1
 
