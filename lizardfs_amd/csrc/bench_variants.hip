@@ -132,10 +132,14 @@ int main(int argc, char **argv) {
 	    {"D4_CH4_swz_nt ", launch_var<4, 4, true, true>, 4, 4, 262144},
 	    {"D4_CH3_swz_nt ", launch_var<4, 3, true, true>, 3, 4, 262144},
 	    {"D4_CH2_swz_nt ", launch_var<4, 2, true, true>, 2, 4, 262144},
+	    {"D4_CH2_ntld   ", launch_var<4, 2, true, true, true>, 2, 4, 262144},
+	    {"D4_CH4_ntld   ", launch_var<4, 4, true, true, true>, 4, 4, 262144},
 	    {"D5_CH2_swz_nt ", launch_var<5, 2, true, true>, 2, 5, 262144},
 	    {"D6_CH4_swz_nt ", launch_var<6, 4, true, true>, 4, 6, 262144},
 	    {"D6_CH3_swz_nt ", launch_var<6, 3, true, true>, 3, 6, 262144},
 	    {"D6_CH2_swz_nt ", launch_var<6, 2, true, true>, 2, 6, 262144},
+	    {"D6_CH2_ntld   ", launch_var<6, 2, true, true, true>, 2, 6, 262144},
+	    {"D6_CH4_ntld   ", launch_var<6, 4, true, true, true>, 4, 6, 262144},
 	    {"D8_CH2_swz_nt ", launch_var<8, 2, true, true>, 2, 8, 262144},
 	    {"D8_CH1_swz_nt ", launch_var<8, 1, true, true>, 1, 8, 262144},
 	};
