@@ -47,7 +47,7 @@ struct Variant {
 	int ch;
 };
 
-template <int D, int CH, bool SWZ, bool NTST, bool NTLD = false>
+template <int D, int CH, bool SWZ, bool NTST, bool NTLD = false, bool TPIPE = false>
 static void launch_var(uint32_t part_len, int srcs, const uint8_t *tbls,
                        const uint64_t *src, const uint64_t *dst, int dests,
                        uint32_t tiles_per_part, uint32_t total_tiles,
@@ -55,7 +55,7 @@ static void launch_var(uint32_t part_len, int srcs, const uint8_t *tbls,
 	uint32_t grid = total_tiles < grid_cap ? total_tiles : grid_cap;
 	size_t lds = (size_t)D * srcs * 32;
 	for (int base = 0; base + D <= dests; base += D)
-		hipLaunchKernelGGL(HIP_KERNEL_NAME(ec_encode_kernel<D, CH, SWZ, NTST, NTLD>),
+		hipLaunchKernelGGL(HIP_KERNEL_NAME(ec_encode_kernel<D, CH, SWZ, NTST, NTLD, TPIPE>),
 		                   dim3(grid), dim3(kThreads), lds, s, part_len, srcs,
 		                   base, tbls, src, dst, dests, tiles_per_part,
 		                   total_tiles);
@@ -126,6 +126,7 @@ int main(int argc, char **argv) {
 	    {"D1_CH4_swz_nt ", launch_var<1, 4, true, true>, 4, 1, 262144},
 	    {"D2_CH4_swz_nt ", launch_var<2, 4, true, true>, 4, 2, 262144},
 	    {"D2_CH4_swz_nt_ntld", launch_var<2, 4, true, true, true>, 4, 2, 262144},
+	    {"D2_CH4_ntld_tp", launch_var<2, 4, true, true, true, true>, 4, 2, 262144},
 	    {"D2_CH6_swz_nt ", launch_var<2, 6, true, true>, 6, 2, 262144},
 	    {"D2_CH4_base   ", launch_var<2, 4, false, false>, 4, 2, 262144},
 	    {"D3_CH4_swz_nt ", launch_var<3, 4, true, true>, 4, 3, 262144},
@@ -133,12 +134,15 @@ int main(int argc, char **argv) {
 	    {"D4_CH3_swz_nt ", launch_var<4, 3, true, true>, 3, 4, 262144},
 	    {"D4_CH2_swz_nt ", launch_var<4, 2, true, true>, 2, 4, 262144},
 	    {"D4_CH2_ntld   ", launch_var<4, 2, true, true, true>, 2, 4, 262144},
+	    {"D4_CH2_ntld_tp", launch_var<4, 2, true, true, true, true>, 2, 4, 262144},
+	    {"D4_CH4_ntld_tp", launch_var<4, 4, true, true, true, true>, 4, 4, 262144},
 	    {"D4_CH4_ntld   ", launch_var<4, 4, true, true, true>, 4, 4, 262144},
 	    {"D5_CH2_swz_nt ", launch_var<5, 2, true, true>, 2, 5, 262144},
 	    {"D6_CH4_swz_nt ", launch_var<6, 4, true, true>, 4, 6, 262144},
 	    {"D6_CH3_swz_nt ", launch_var<6, 3, true, true>, 3, 6, 262144},
 	    {"D6_CH2_swz_nt ", launch_var<6, 2, true, true>, 2, 6, 262144},
 	    {"D6_CH2_ntld   ", launch_var<6, 2, true, true, true>, 2, 6, 262144},
+	    {"D6_CH2_ntld_tp", launch_var<6, 2, true, true, true, true>, 2, 6, 262144},
 	    {"D6_CH4_ntld   ", launch_var<6, 4, true, true, true>, 4, 6, 262144},
 	    {"D8_CH2_swz_nt ", launch_var<8, 2, true, true>, 2, 8, 262144},
 	    {"D8_CH1_swz_nt ", launch_var<8, 1, true, true>, 1, 8, 262144},

@@ -72,7 +72,8 @@ __device__ __forceinline__ uint4 ld_nt(const uint8_t *p) {
 	return make_uint4(v.x, v.y, v.z, v.w);
 }
 
-template <int D, int CH, bool SWZ, bool NTST, bool NTLD = false>
+template <int D, int CH, bool SWZ, bool NTST, bool NTLD = false,
+          bool TPIPE = false>
 __global__ __launch_bounds__(kThreads) void ec_encode_kernel(
     uint32_t part_len, int srcs, int dest_base,
     const uint8_t *__restrict__ gftbls_dev,  /* 32*srcs*dests_total */
@@ -109,9 +110,20 @@ __global__ __launch_bounds__(kThreads) void ec_encode_kernel(
 				acc[d][c] = make_uint4(0, 0, 0, 0);
 
 		if ((uint64_t)(tin + 1) * kTile <= part_len) {
-			/* full tile: branchless, next source prefetched while the
-			 * current one is accumulated */
+			/* full tile: branchless; next source's strips — and, with
+			 * TPIPE, its coefficient tables — prefetched while the
+			 * current source is accumulated (the per-source LDS
+			 * broadcast + lgkmcnt wait otherwise serializes large k) */
 			uint4 w[CH], wn[CH];
+			uint4 L[TPIPE ? 2 : 1][D], H[TPIPE ? 2 : 1][D];
+			auto tbl_read = [&](int j, uint4 (&Lb)[D], uint4 (&Hb)[D]) {
+#pragma unroll
+				for (int d = 0; d < D; ++d) {
+					const uint8_t *tb = smem + ((size_t)d * srcs + j) * 32;
+					Lb[d] = *(const uint4 *)tb;
+					Hb[d] = *(const uint4 *)(tb + 16);
+				}
+			};
 			{
 				const uint8_t *sp = (const uint8_t *)sp_tab[0];
 #pragma unroll
@@ -119,6 +131,7 @@ __global__ __launch_bounds__(kThreads) void ec_encode_kernel(
 					w[c] = NTLD ? ld_nt(sp + (base + c * kChunkBytes))
 					            : *(const uint4 *)(sp + (base + c * kChunkBytes));
 			}
+			if (TPIPE) tbl_read(0, L[0], H[0]);
 			for (int j = 0; j < srcs; ++j) {
 				if (j + 1 < srcs) {
 					const uint8_t *spn = (const uint8_t *)sp_tab[j + 1];
@@ -127,16 +140,17 @@ __global__ __launch_bounds__(kThreads) void ec_encode_kernel(
 						wn[c] = NTLD ? ld_nt(spn + (base + c * kChunkBytes))
 						             : *(const uint4 *)(spn + (base + c * kChunkBytes));
 				}
-				uint4 L[D], H[D];
-#pragma unroll
-				for (int d = 0; d < D; ++d) {
-					const uint8_t *tb = smem + ((size_t)d * srcs + j) * 32;
-					L[d] = *(const uint4 *)tb;
-					H[d] = *(const uint4 *)(tb + 16);
+				int cur = 0;
+				if (TPIPE) {
+					cur = j & 1;
+					if (j + 1 < srcs)
+						tbl_read(j + 1, L[(j + 1) & 1], H[(j + 1) & 1]);
+				} else {
+					tbl_read(j, L[0], H[0]);
 				}
 #pragma unroll
 				for (int c = 0; c < CH; ++c)
-					gf_macc_all<D, CH>(acc, c, w[c], L, H);
+					gf_macc_all<D, CH>(acc, c, w[c], L[cur], H[cur]);
 #pragma unroll
 				for (int c = 0; c < CH; ++c) w[c] = wn[c];
 			}
