@@ -131,29 +131,34 @@ __global__ __launch_bounds__(kThreads) void ec_encode_kernel(
 					w[c] = NTLD ? ld_nt(sp + (base + c * kChunkBytes))
 					            : *(const uint4 *)(sp + (base + c * kChunkBytes));
 			}
-			if (TPIPE) tbl_read(0, L[0], H[0]);
-			for (int j = 0; j < srcs; ++j) {
+			/* one source step: compute with (Lc,Hc) while prefetching the
+			 * next source's strips and (with TPIPE) its tables into
+			 * (Ln,Hn) — all buffer indices static so everything stays in
+			 * registers */
+			auto step = [&](int j, uint4 (&Lc)[D], uint4 (&Hc)[D],
+			                uint4 (&Ln)[D], uint4 (&Hn)[D]) {
 				if (j + 1 < srcs) {
 					const uint8_t *spn = (const uint8_t *)sp_tab[j + 1];
 #pragma unroll
 					for (int c = 0; c < CH; ++c)
 						wn[c] = NTLD ? ld_nt(spn + (base + c * kChunkBytes))
 						             : *(const uint4 *)(spn + (base + c * kChunkBytes));
+					if (TPIPE) tbl_read(j + 1, Ln, Hn);
 				}
-				int cur = 0;
-				if (TPIPE) {
-					cur = j & 1;
-					if (j + 1 < srcs)
-						tbl_read(j + 1, L[(j + 1) & 1], H[(j + 1) & 1]);
-				} else {
-					tbl_read(j, L[0], H[0]);
-				}
+				if (!TPIPE) tbl_read(j, Lc, Hc);
 #pragma unroll
 				for (int c = 0; c < CH; ++c)
-					gf_macc_all<D, CH>(acc, c, w[c], L[cur], H[cur]);
+					gf_macc_all<D, CH>(acc, c, w[c], Lc, Hc);
 #pragma unroll
 				for (int c = 0; c < CH; ++c) w[c] = wn[c];
+			};
+			if (TPIPE) tbl_read(0, L[0], H[0]);
+			int j = 0;
+			for (; j + 1 < srcs; j += 2) {
+				step(j, L[0], H[0], L[TPIPE ? 1 : 0], H[TPIPE ? 1 : 0]);
+				step(j + 1, L[TPIPE ? 1 : 0], H[TPIPE ? 1 : 0], L[0], H[0]);
 			}
+			if (j < srcs) step(j, L[0], H[0], L[TPIPE ? 1 : 0], H[TPIPE ? 1 : 0]);
 #pragma unroll
 			for (int d = 0; d < D; ++d) {
 				uint8_t *dp = (uint8_t *)dp_tab[d];
