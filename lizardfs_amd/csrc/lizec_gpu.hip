@@ -55,7 +55,7 @@ constexpr int ec_chunks_for(int d) { return d <= 2 ? 4 : 2; }
  *  [0..4*256)      u32: slicing tables T0..T3 (reflected, poly 0xEDB88320)
  *  [1024..1024+26*32) u32: advance matrices M_i = "append 2^i zero BYTES",
  *                      i = 0..25 (supports block_len < 64 MiB)            */
-constexpr int kCrcTabWords = 8 * 256;   /* slicing-by-8 tables T0..T7 */
+constexpr int kCrcTabWords = 16 * 256;  /* slicing-by-16 tables T0..T15 */
 constexpr int kCrcMatCount = 26;
 constexpr int kCrcConstWords = kCrcTabWords + kCrcMatCount * 32;
 
@@ -167,25 +167,33 @@ __device__ uint32_t crc_block_wave(const uint8_t *__restrict__ block,
 #pragma unroll
 			for (int q = 0; q < BV; ++q)
 				w[c][q] = *(const uint4 *)(base + c * span + i + q * 16);
-#define LIZEC_CRC8(crc, lo, hi)                                          \
+/* one 16-byte step: 16 independent lookups per chain (slicing-by-16 —
+ * the serial dependency is one LDS round trip per 16 bytes) */
+#define LIZEC_CRC16(crc, v)                                              \
 	do {                                                                 \
-		uint32_t u0 = (crc) ^ (lo), u1 = (hi);                           \
-		(crc) = T[7 * 256 + (u0 & 0xff)] ^                               \
-		        T[6 * 256 + ((u0 >> 8) & 0xff)] ^                        \
-		        T[5 * 256 + ((u0 >> 16) & 0xff)] ^                       \
-		        T[4 * 256 + (u0 >> 24)] ^                                \
-		        T[3 * 256 + (u1 & 0xff)] ^                               \
-		        T[2 * 256 + ((u1 >> 8) & 0xff)] ^                        \
-		        T[1 * 256 + ((u1 >> 16) & 0xff)] ^ T[u1 >> 24];          \
+		uint32_t u0 = (crc) ^ (v).x, u1 = (v).y, u2 = (v).z, u3 = (v).w; \
+		(crc) = T[15 * 256 + (u0 & 0xff)] ^                              \
+		        T[14 * 256 + ((u0 >> 8) & 0xff)] ^                       \
+		        T[13 * 256 + ((u0 >> 16) & 0xff)] ^                      \
+		        T[12 * 256 + (u0 >> 24)] ^                               \
+		        T[11 * 256 + (u1 & 0xff)] ^                              \
+		        T[10 * 256 + ((u1 >> 8) & 0xff)] ^                       \
+		        T[9 * 256 + ((u1 >> 16) & 0xff)] ^                       \
+		        T[8 * 256 + (u1 >> 24)] ^                                \
+		        T[7 * 256 + (u2 & 0xff)] ^                               \
+		        T[6 * 256 + ((u2 >> 8) & 0xff)] ^                        \
+		        T[5 * 256 + ((u2 >> 16) & 0xff)] ^                       \
+		        T[4 * 256 + (u2 >> 24)] ^                                \
+		        T[3 * 256 + (u3 & 0xff)] ^                               \
+		        T[2 * 256 + ((u3 >> 8) & 0xff)] ^                        \
+		        T[1 * 256 + ((u3 >> 16) & 0xff)] ^ T[u3 >> 24];          \
 	} while (0)
 #pragma unroll
 		for (int q = 0; q < BV; ++q)
 #pragma unroll
-			for (int c = 0; c < C; ++c) {
-				LIZEC_CRC8(crc[c], w[c][q].x, w[c][q].y);
-				LIZEC_CRC8(crc[c], w[c][q].z, w[c][q].w);
-			}
-#undef LIZEC_CRC8
+			for (int c = 0; c < C; ++c)
+				LIZEC_CRC16(crc[c], w[c][q]);
+#undef LIZEC_CRC16
 	}
 #pragma unroll
 	for (int c = 0; c < C; ++c) crc[c] ^= 0xFFFFFFFFu;
@@ -311,7 +319,7 @@ static void build_crc_const(uint32_t *w) {
 		w[i] = c;
 	}
 	for (uint32_t i = 0; i < 256; ++i)
-		for (int t = 1; t < 8; ++t) {
+		for (int t = 1; t < 16; ++t) {
 			uint32_t c = w[(t - 1) * 256 + i];
 			w[t * 256 + i] = w[c & 0xff] ^ (c >> 8);
 		}
