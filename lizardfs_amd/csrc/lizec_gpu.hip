@@ -148,7 +148,7 @@ __global__ __launch_bounds__(kThreads) void crc32_blocks_kernel(
 /* Per-wave CRC of one block: C chains x 64 lane segments, BV*16-byte
  * bursts, slicing-by-8; returns the block CRC on every lane (lane 0's
  * value is authoritative). */
-template <int C, int BV>
+template <int C, int BV, int SL = 16>
 __device__ uint32_t crc_block_wave(const uint8_t *__restrict__ block,
                                    uint32_t block_len, uint32_t seed,
                                    const uint32_t *T, const uint32_t *mats,
@@ -188,12 +188,30 @@ __device__ uint32_t crc_block_wave(const uint8_t *__restrict__ block,
 		        T[2 * 256 + ((u3 >> 8) & 0xff)] ^                        \
 		        T[1 * 256 + ((u3 >> 16) & 0xff)] ^ T[u3 >> 24];          \
 	} while (0)
+#define LIZEC_CRC8(crc, lo, hi)                                          \
+	do {                                                                 \
+		uint32_t u0 = (crc) ^ (lo), u1 = (hi);                           \
+		(crc) = T[7 * 256 + (u0 & 0xff)] ^                               \
+		        T[6 * 256 + ((u0 >> 8) & 0xff)] ^                        \
+		        T[5 * 256 + ((u0 >> 16) & 0xff)] ^                       \
+		        T[4 * 256 + (u0 >> 24)] ^                                \
+		        T[3 * 256 + (u1 & 0xff)] ^                               \
+		        T[2 * 256 + ((u1 >> 8) & 0xff)] ^                        \
+		        T[1 * 256 + ((u1 >> 16) & 0xff)] ^ T[u1 >> 24];          \
+	} while (0)
 #pragma unroll
 		for (int q = 0; q < BV; ++q)
 #pragma unroll
-			for (int c = 0; c < C; ++c)
-				LIZEC_CRC16(crc[c], w[c][q]);
+			for (int c = 0; c < C; ++c) {
+				if (SL == 16) {
+					LIZEC_CRC16(crc[c], w[c][q]);
+				} else {
+					LIZEC_CRC8(crc[c], w[c][q].x, w[c][q].y);
+					LIZEC_CRC8(crc[c], w[c][q].z, w[c][q].w);
+				}
+			}
 #undef LIZEC_CRC16
+#undef LIZEC_CRC8
 	}
 #pragma unroll
 	for (int c = 0; c < C; ++c) crc[c] ^= 0xFFFFFFFFu;
@@ -216,7 +234,7 @@ __device__ uint32_t crc_block_wave(const uint8_t *__restrict__ block,
 	return acc;
 }
 
-template <int C, int BV>
+template <int C, int BV, int SL = 16>
 __global__ __launch_bounds__(kThreads) void crc32_blocks_kernel_multi(
     const uint8_t *__restrict__ buf, uint32_t block_len, uint64_t nblocks,
     uint32_t seed, const uint32_t *__restrict__ crc_const,
@@ -231,8 +249,9 @@ __global__ __launch_bounds__(kThreads) void crc32_blocks_kernel_multi(
 	const int lane = threadIdx.x & 63;
 	for (uint64_t blk = (uint64_t)blockIdx.x * 4 + wave; blk < nblocks;
 	     blk += (uint64_t)gridDim.x * 4) {
-		uint32_t crc = crc_block_wave<C, BV>(buf + blk * block_len, block_len,
-		                                     seed, T, mats, lane);
+		uint32_t crc = crc_block_wave<C, BV, SL>(buf + blk * block_len,
+		                                         block_len, seed, T, mats,
+		                                         lane);
 		if (lane == 0) out[blk] = crc;
 	}
 }
@@ -566,10 +585,17 @@ extern "C" int lizec_crc32_batch(lizec_engine *e, const void *dev_buf,
 	LIZEC_CHECK(hipSetDevice(e->device));
 	uint64_t groups = (nblocks + 3) / 4;
 	uint32_t grid = (uint32_t)(groups < 131072 ? groups : 131072);
-	const char *ch = getenv("LIZEC_CRC_CHAINS");   /* A/B hook */
+	const char *ch = getenv("LIZEC_CRC_CHAINS");   /* A/B hooks */
+	const char *sl = getenv("LIZEC_CRC_SLICE");
 	int chains = ch ? atoi(ch) : 2;   /* C=4 measured worse: half-line bursts re-thrash L1 (profiles) */
+	int slice = sl ? atoi(sl) : 16;
 	if (block_len % 32768 == 0 && chains >= 4)
 		hipLaunchKernelGGL(HIP_KERNEL_NAME(crc32_blocks_kernel_multi<4, 8>),
+		                   dim3(grid), dim3(kThreads), 0, s,
+		                   (const uint8_t *)dev_buf, block_len, nblocks,
+		                   seed, e->d_crc_const, dev_crcs_out);
+	else if (block_len % 16384 == 0 && slice == 8)
+		hipLaunchKernelGGL(HIP_KERNEL_NAME(crc32_blocks_kernel_multi<2, 8, 8>),
 		                   dim3(grid), dim3(kThreads), 0, s,
 		                   (const uint8_t *)dev_buf, block_len, nblocks,
 		                   seed, e->d_crc_const, dev_crcs_out);
