@@ -135,6 +135,29 @@ extern "C" void ec_encode_data(int len, int srcs, int dests, uint8_t *v,
 	}
 }
 
+
+/* distinct-name C entry points for the C++-mangled alias TU
+ * (lizec_abi_aliases.cpp) — avoids linkage conflicts in one TU. */
+extern "C" void lizec_impl_gen_rs_matrix(uint8_t *a, int m, int k) {
+	gf_gen_rs_matrix(a, m, k);
+}
+extern "C" void lizec_impl_gen_cauchy1_matrix(uint8_t *a, int m, int k) {
+	gf_gen_cauchy1_matrix(a, m, k);
+}
+extern "C" int lizec_impl_invert_matrix(uint8_t *in_mat, uint8_t *out_mat,
+                                        int n) {
+	return gf_invert_matrix(in_mat, out_mat, n);
+}
+extern "C" void lizec_impl_init_tables(int k, int rows, uint8_t *a,
+                                       uint8_t *g_tbls) {
+	ec_init_tables(k, rows, a, g_tbls);
+}
+extern "C" void lizec_impl_encode_data(int len, int srcs, int dests,
+                                       uint8_t *v, uint8_t **src,
+                                       uint8_t **dest) {
+	ec_encode_data(len, srcs, dests, v, src, dest);
+}
+
 /* ------------------------------------------------------------------ */
 /* ReedSolomon table builders (reed_solomon.h:41-373 semantics)       */
 /* ------------------------------------------------------------------ */
