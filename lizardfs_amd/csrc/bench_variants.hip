@@ -39,14 +39,6 @@ __global__ void fill_kernel(uint8_t *p, size_t n, uint32_t salt) {
 	}
 }
 
-struct Variant {
-	const char *name;
-	void (*launch)(uint32_t, int, const uint8_t *, const uint64_t *,
-	               const uint64_t *, int, uint32_t, uint32_t, uint32_t,
-	               hipStream_t);
-	int ch;
-};
-
 template <int D, int CH, bool SWZ, bool NTST, bool NTLD = false, bool TPIPE = false>
 static void launch_var(uint32_t part_len, int srcs, const uint8_t *tbls,
                        const uint64_t *src, const uint64_t *dst, int dests,

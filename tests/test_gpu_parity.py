@@ -189,11 +189,14 @@ def test_crc_blocks_matches_golden_and_oracle(crc_mod):
     got = crc_mod.crc32_blocks(to_gpu(buf), 65536).cpu().numpy().view(np.uint32)
     exp = oracle.crc32_blocks(buf, 65536)
     assert np.array_equal(got, exp)
-    # different block size + seed
-    got = crc_mod.crc32_blocks(to_gpu(buf), 1024, seed=0xABCD1234) \
-        .cpu().numpy().view(np.uint32)
-    exp = oracle.crc32_blocks(buf, 1024, seed=0xABCD1234)
-    assert np.array_equal(got, exp)
+    # other block sizes + seed: generic kernel (1 KiB, 2 KiB) and the
+    # fast kernel off the 64 KiB shape (48 KiB)
+    for blen in (1024, 2048, 49152):
+        got = crc_mod.crc32_blocks(to_gpu(buf[:8 * 49152]), blen,
+                                   seed=0xABCD1234) \
+            .cpu().numpy().view(np.uint32)
+        exp = oracle.crc32_blocks(buf[:8 * 49152], blen, seed=0xABCD1234)
+        assert np.array_equal(got, exp), blen
     # all-zero blocks (sparse-chunk special case, crc.cc:235-243 context)
     z = np.zeros(4 * 65536, np.uint8)
     got = crc_mod.crc32_blocks(to_gpu(z), 65536).cpu().numpy().view(np.uint32)
