@@ -211,6 +211,18 @@ int lizec_scrub_batch(lizec_engine *e, const uint64_t *chunk_dptrs,
                       const uint32_t *block_counts, int nchunks,
                       int32_t *dev_status_out, void *stream);
 
+/* Generalized layout: block b's data at data_offs[i] + b*block_stride,
+ * its stored CRC at crc_offs[i] + b*crc_stride.  Covers both on-disk
+ * formats: MooseFS (stride 65536 / 4; the wrapper above) and INTERLEAVED
+ * (kHddBlockSize = 65540, chunk.h:40: 4-byte CRC inline before each
+ * block — data_off 4, both strides 65540, crc_off 0). */
+int lizec_scrub_batch_strided(lizec_engine *e, const uint64_t *chunk_dptrs,
+                              const uint32_t *data_offs,
+                              const uint32_t *crc_offs,
+                              const uint32_t *block_counts, int nchunks,
+                              uint32_t block_stride, uint32_t crc_stride,
+                              int32_t *dev_status_out, void *stream);
+
 #ifdef __cplusplus
 }
 #endif

@@ -268,8 +268,8 @@ __global__ __launch_bounds__(kThreads) void scrub_chunks_kernel(
     const uint32_t *__restrict__ data_offs,
     const uint32_t *__restrict__ crc_offs,
     const uint32_t *__restrict__ block_counts, uint32_t nchunks,
-    uint32_t max_blocks, const uint32_t *__restrict__ crc_const,
-    int32_t *__restrict__ status) {
+    uint32_t max_blocks, uint32_t block_stride, uint32_t crc_stride,
+    const uint32_t *__restrict__ crc_const, int32_t *__restrict__ status) {
 	__shared__ __attribute__((aligned(16))) uint32_t stabs[kCrcConstWords];
 	for (int i = threadIdx.x; i < kCrcConstWords; i += kThreads)
 		stabs[i] = crc_const[i];
@@ -285,10 +285,10 @@ __global__ __launch_bounds__(kThreads) void scrub_chunks_kernel(
 		uint32_t b = (uint32_t)(flat - (uint64_t)c * max_blocks);
 		if (b >= block_counts[c]) continue;
 		const uint8_t *img = (const uint8_t *)chunk_dptrs[c];
-		uint32_t crc = crc_block_wave<2, 8>(img + data_offs[c] + b * 65536u,
-		                                    65536u, 0u, T, mats, lane);
+		uint32_t crc = crc_block_wave<2, 8, 8>(
+		    img + data_offs[c] + b * block_stride, 65536u, 0u, T, mats, lane);
 		if (lane == 0) {
-			const uint8_t *p = img + crc_offs[c] + 4u * b;
+			const uint8_t *p = img + crc_offs[c] + crc_stride * b;
 			uint32_t stored = ((uint32_t)p[0] << 24) | ((uint32_t)p[1] << 16) |
 			                  ((uint32_t)p[2] << 8) | p[3];
 			if (stored != crc) atomicMin(&status[c], (int32_t)b);
@@ -612,11 +612,11 @@ extern "C" int lizec_crc32_batch(lizec_engine *e, const void *dev_buf,
 	return LIZEC_OK;
 }
 
-extern "C" int lizec_scrub_batch(lizec_engine *e, const uint64_t *chunk_dptrs,
-                                 const uint32_t *data_offs,
-                                 const uint32_t *crc_offs,
-                                 const uint32_t *block_counts, int nchunks,
-                                 int32_t *dev_status_out, void *stream) {
+extern "C" int lizec_scrub_batch_strided(
+    lizec_engine *e, const uint64_t *chunk_dptrs, const uint32_t *data_offs,
+    const uint32_t *crc_offs, const uint32_t *block_counts, int nchunks,
+    uint32_t block_stride, uint32_t crc_stride, int32_t *dev_status_out,
+    void *stream) {
 	if (!e || !chunk_dptrs || nchunks < 1 || !dev_status_out)
 		return LIZEC_EINVAL;
 	hipStream_t s = stream ? (hipStream_t)stream : e->stream;
@@ -650,7 +650,19 @@ extern "C" int lizec_scrub_batch(lizec_engine *e, const uint64_t *chunk_dptrs,
 	uint32_t grid = (uint32_t)(groups < 131072 ? groups : 131072);
 	hipLaunchKernelGGL(scrub_chunks_kernel, dim3(grid), dim3(kThreads), 0, s,
 	                   d_ptrs, d_doffs, d_coffs, d_counts, (uint32_t)nchunks,
-	                   max_blocks, e->d_crc_const, dev_status_out);
+	                   max_blocks, block_stride, crc_stride, e->d_crc_const,
+	                   dev_status_out);
 	LIZEC_CHECK(hipGetLastError());
 	return LIZEC_OK;
+}
+
+/* MooseFS layout: 64 KiB blocks after the header, 4-byte CRC array. */
+extern "C" int lizec_scrub_batch(lizec_engine *e, const uint64_t *chunk_dptrs,
+                                 const uint32_t *data_offs,
+                                 const uint32_t *crc_offs,
+                                 const uint32_t *block_counts, int nchunks,
+                                 int32_t *dev_status_out, void *stream) {
+	return lizec_scrub_batch_strided(e, chunk_dptrs, data_offs, crc_offs,
+	                                 block_counts, nchunks, 65536u, 4u,
+	                                 dev_status_out, stream);
 }

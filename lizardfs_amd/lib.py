@@ -91,6 +91,12 @@ def lib():
         ctypes.POINTER(ctypes.c_uint32), ctypes.POINTER(ctypes.c_uint32),
         ctypes.POINTER(ctypes.c_uint32), ctypes.c_int, ctypes.c_void_p,
         ctypes.c_void_p]
+    L.lizec_scrub_batch_strided.restype = ctypes.c_int
+    L.lizec_scrub_batch_strided.argtypes = [
+        ctypes.c_void_p, ctypes.POINTER(ctypes.c_uint64),
+        ctypes.POINTER(ctypes.c_uint32), ctypes.POINTER(ctypes.c_uint32),
+        ctypes.POINTER(ctypes.c_uint32), ctypes.c_int, ctypes.c_uint32,
+        ctypes.c_uint32, ctypes.c_void_p, ctypes.c_void_p]
     _lib = L
     return _lib
 

@@ -22,6 +22,7 @@ from . import slice_traits as st
 SIGNATURE_ID = b"LIZC 1.1"        # chunk_signature.cc:30
 SIGNATURE_BLOCK = 1024            # chunk.h:156 kMaxSignatureBlockSize
 DISK_BLOCK = 4096                 # chunk.h:161 kDiskBlockSize
+HDD_BLOCK = st.BLOCK_SIZE + 4     # chunk.h:40 kHddBlockSize (interleaved)
 CLEAN = 0x7FFFFFFF
 
 
@@ -56,15 +57,26 @@ def parse_signature(buf):
 
 
 def build_chunk_image(chunk_id, version, slice_type, part, blocks_data,
-                      crc32_fn=None):
-    """Assemble a MooseFS part-file image (numpy uint8) from 64 KiB blocks.
-    Used by tests/bench; the CRC array is filled with crc32_fn (default:
-    liblizec host crc32)."""
+                      crc32_fn=None, fmt="moosefs"):
+    """Assemble a part-file image (numpy uint8) from 64 KiB blocks in
+    either on-disk format (chunk.cc MooseFSChunk / InterleavedChunk).
+    Used by tests/bench; CRCs are filled with crc32_fn (default: liblizec
+    host crc32) and stored big-endian (put32bit)."""
     if crc32_fn is None:
         from . import crc as lcrc
         crc32_fn = lcrc.crc32
-    hdr = header_size(slice_type)
     nblocks = len(blocks_data)
+    if fmt == "interleaved":
+        # chunk.cc:195-209: no signature; 4-byte CRC inline before each block
+        img = np.zeros(nblocks * HDD_BLOCK, np.uint8)
+        for b, blk in enumerate(blocks_data):
+            assert blk.size == st.BLOCK_SIZE
+            crc = crc32_fn(blk.tobytes())
+            img[b * HDD_BLOCK:b * HDD_BLOCK + 4] = \
+                np.frombuffer(struct.pack(">I", crc), np.uint8)
+            img[b * HDD_BLOCK + 4:(b + 1) * HDD_BLOCK] = blk
+        return img
+    hdr = header_size(slice_type)
     assert nblocks <= max_blocks_in_file(slice_type)
     img = np.zeros(hdr + nblocks * st.BLOCK_SIZE, np.uint8)
     sig = build_signature(chunk_id, version, slice_type, part)
@@ -78,59 +90,75 @@ def build_chunk_image(chunk_id, version, slice_type, part, blocks_data,
     return img
 
 
-def scrub_batch(images, device=0):
-    """Verify a batch of chunk-part images resident on the GPU.
-
-    images: list of (tensor, slice_type) — tensor is the flat uint8 CUDA
-    image (full file bytes).  Validates each signature (host-side, 1 KiB
-    read, as hdd_int_chunk_readcrc does) then CRC-checks every block on
-    the GPU.  Returns list of first-damaged-block index or None if clean;
-    a bad signature reports -2 (the reference's LIZARDFS_ERROR_IO class).
-    """
-    n = len(images)
+def _scrub_group(idx, images, fmts, device, results):
+    """One lizec_scrub_batch_strided call for images sharing a format."""
+    n = len(idx)
     dptrs = np.zeros(n, np.uint64)
     doffs = np.zeros(n, np.uint32)
     coffs = np.zeros(n, np.uint32)
     counts = np.zeros(n, np.uint32)
-    sig_bad = [False] * n
-    for i, (img, slice_type) in enumerate(images):
-        if img.dtype != torch.uint8 or not img.is_cuda or \
-                not img.is_contiguous():
-            raise ValueError("images must be contiguous CUDA uint8 tensors")
-        head = img[:SIGNATURE_BLOCK].cpu().numpy()
-        parsed = parse_signature(head)
-        if parsed is None or parsed[2] != slice_type:
-            sig_bad[i] = True
-            counts[i] = 0
-            continue
-        hdr = header_size(slice_type)
-        nb = (img.numel() - hdr) // st.BLOCK_SIZE
-        dptrs[i] = img.data_ptr()
-        doffs[i] = hdr
-        coffs[i] = SIGNATURE_BLOCK
-        counts[i] = nb
+    fmt = fmts[idx[0]]
+    if fmt == "interleaved":
+        bstride, cstride = HDD_BLOCK, HDD_BLOCK
+    else:
+        bstride, cstride = st.BLOCK_SIZE, 4
+    for j, i in enumerate(idx):
+        img, slice_type = images[i][0], images[i][1]
+        if fmt == "interleaved":
+            dptrs[j] = img.data_ptr()
+            doffs[j] = 4
+            coffs[j] = 0
+            counts[j] = img.numel() // HDD_BLOCK
+        else:
+            hdr = header_size(slice_type)
+            dptrs[j] = img.data_ptr()
+            doffs[j] = hdr
+            coffs[j] = SIGNATURE_BLOCK
+            counts[j] = (img.numel() - hdr) // st.BLOCK_SIZE
     if counts.max(initial=0) == 0:
-        return [-2 if bad else None for bad in sig_bad]
-
+        return
     status = torch.empty(n, dtype=torch.int32, device=f"cuda:{device}")
     stream = torch.cuda.current_stream(device).cuda_stream
-    # chunks with bad signatures get count 0 (skipped; stay CLEAN on GPU)
-    counts_safe = np.maximum(counts, 0)
-    L.check(L.lib().lizec_scrub_batch(
+    L.check(L.lib().lizec_scrub_batch_strided(
         L.engine(device),
         dptrs.ctypes.data_as(ctypes.POINTER(ctypes.c_uint64)),
         doffs.ctypes.data_as(ctypes.POINTER(ctypes.c_uint32)),
         coffs.ctypes.data_as(ctypes.POINTER(ctypes.c_uint32)),
-        counts_safe.ctypes.data_as(ctypes.POINTER(ctypes.c_uint32)),
-        n, ctypes.c_void_p(status.data_ptr()), ctypes.c_void_p(stream)),
-        "lizec_scrub_batch")
+        counts.ctypes.data_as(ctypes.POINTER(ctypes.c_uint32)),
+        n, bstride, cstride, ctypes.c_void_p(status.data_ptr()),
+        ctypes.c_void_p(stream)), "lizec_scrub_batch_strided")
     torch.cuda.synchronize(device)
-    out = []
-    for i, s in enumerate(status.cpu().numpy()):
-        if sig_bad[i]:
-            out.append(-2)
-        elif s == CLEAN:
-            out.append(None)
-        else:
-            out.append(int(s))
-    return out
+    for j, i in enumerate(idx):
+        s = int(status[j].item())
+        results[i] = None if s == CLEAN else s
+
+
+def scrub_batch(images, device=0):
+    """Verify a batch of chunk-part images resident on the GPU.
+
+    images: list of (tensor, slice_type) or (tensor, slice_type, fmt) with
+    fmt in {"moosefs" (default), "interleaved"} — tensor is the flat uint8
+    CUDA image (full file bytes).  MooseFS images get a host-side signature
+    check (1 KiB read, as hdd_int_chunk_readcrc does); then every block is
+    CRC-checked on the GPU.  Returns list of first-damaged-block index or
+    None if clean; a bad signature reports -2 (the reference's
+    LIZARDFS_ERROR_IO class)."""
+    n = len(images)
+    images = [t if len(t) == 3 else (t[0], t[1], "moosefs") for t in images]
+    fmts = [t[2] for t in images]
+    results = [None] * n
+    groups = {}
+    for i, (img, slice_type, fmt) in enumerate(images):
+        if img.dtype != torch.uint8 or not img.is_cuda or \
+                not img.is_contiguous():
+            raise ValueError("images must be contiguous CUDA uint8 tensors")
+        if fmt == "moosefs":
+            head = img[:SIGNATURE_BLOCK].cpu().numpy()
+            parsed = parse_signature(head)
+            if parsed is None or parsed[2] != slice_type:
+                results[i] = -2
+                continue
+        groups.setdefault(fmt, []).append(i)
+    for fmt, idx in groups.items():
+        _scrub_group(idx, images, fmts, device, results)
+    return results

@@ -58,6 +58,38 @@ def test_build_image_crc_matches_oracle():
 
 
 @pytest.mark.gpu
+def test_scrub_interleaved_gpu():
+    """INTERLEAVED on-disk format (chunk.h:40 kHddBlockSize = 65540: 4-byte
+    BE CRC inline before each 64 KiB block, no signature block)."""
+    import torch
+    rng = np.random.default_rng(8)
+    t = st.ec_slice_type(8, 2)
+
+    def make(nblocks):
+        blocks = [rng.integers(0, 256, st.BLOCK_SIZE, np.uint8)
+                  for _ in range(nblocks)]
+        return scrub.build_chunk_image(7, 1, t, 0, blocks,
+                                       crc32_fn=lambda b: oracle.crc32(b),
+                                       fmt="interleaved")
+
+    clean = make(3)
+    bad = make(4)
+    bad[2 * 65540 + 4 + 100] ^= 0x10   # corrupt block 2's data
+    badcrc = make(2)
+    badcrc[1 * 65540 + 3] ^= 0x01      # corrupt block 1's stored CRC
+    batch = [(torch.from_numpy(x).cuda(), t, "interleaved")
+             for x in (clean, bad, badcrc)]
+    # mix formats in one call
+    moose = scrub.build_chunk_image(9, 1, t, 1,
+                                    [rng.integers(0, 256, st.BLOCK_SIZE,
+                                                  np.uint8)],
+                                    crc32_fn=lambda b: oracle.crc32(b))
+    batch.append((torch.from_numpy(moose).cuda(), t))
+    res = scrub.scrub_batch(batch)
+    assert res == [None, 2, 1, None], res
+
+
+@pytest.mark.gpu
 def test_scrub_batch_gpu():
     import torch
     rng = np.random.default_rng(6)
