@@ -72,6 +72,79 @@ int main() {
 """
 
 
+RS_PROBE = r"""
+#include <cstdio>
+#include <cstdint>
+#include <cstring>
+#include <stdexcept>
+/* ISA-L mode: reed_solomon.h:27-31 pulls isa-l/erasure_code.h — served by
+ * liblizec's shim (include/isa-l/erasure_code.h). */
+#define LIZARDFS_HAVE_ISA_L_ERASURE_CODE_H 1
+#include "common/reed_solomon.h"
+
+int main() {
+	const int k = 8, m = 2, n = 512;
+	typedef ReedSolomon<32, 32> RS;
+	RS rs(k, m);
+	uint8_t data[8][512], par[2][512], rec[2][512];
+	for (int j = 0; j < k; ++j)
+		for (int i = 0; i < n; ++i) data[j][i] = (uint8_t)(j * 53 + i * 7);
+
+	RS::ConstFragmentMap dmap{{0}};
+	RS::FragmentMap pmap{{0}};
+	for (int j = 0; j < k; ++j) dmap[j] = data[j];
+	for (int l = 0; l < m; ++l) pmap[l] = par[l];
+	rs.encode(dmap, pmap, n);
+	for (int l = 0; l < m; ++l)
+		for (int i = 0; i < n; ++i) printf("%02x", par[l][i]);
+	printf("\n");
+
+	/* erase data parts 1 and 5, recover through the reference's recover() */
+	RS::ConstFragmentMap in{{0}};
+	RS::FragmentMap out{{0}};
+	RS::ErasedMap erased;
+	for (int i = 0; i < k + m; ++i) {
+		if (i == 1 || i == 5) { erased.set(i); continue; }
+		in[i] = i < k ? data[i] : par[i - k];
+	}
+	out[1] = rec[0];
+	out[5] = rec[1];
+	rs.recover(in, erased, out, n);
+	int ok1 = memcmp(rec[0], data[1], n) == 0;
+	int ok2 = memcmp(rec[1], data[5], n) == 0;
+	printf("%d %d\n", ok1, ok2);
+	return 0;
+}
+"""
+
+
+def test_reference_reed_solomon_runs_on_liblizec(tmp_path):
+    """Compile the REFERENCE'S ReedSolomon<32,32> (reed_solomon.h) in its
+    ISA-L configuration against liblizec's shim + exports, and run the
+    reference's own encode() and recover() on our backend."""
+    src = tmp_path / "rs_probe.cc"
+    src.write_text(RS_PROBE)
+    exe = tmp_path / "rs_probe"
+    lib = os.path.join(REPO, "lizardfs_amd", "liblizec.so")
+    cmd = ["g++", "-O1", "-std=c++14", str(src), "-o", str(exe),
+           f"-I{REF}/src", f"-I{REPO}/include",
+           f"-I{REPO}/oracle/ref_config",
+           lib, f"-Wl,-rpath,{os.path.dirname(lib)}"]
+    r = subprocess.run(cmd, capture_output=True, text=True)
+    assert r.returncode == 0, r.stderr
+    out = subprocess.run([str(exe)], capture_output=True, text=True)
+    assert out.returncode == 0, out.stderr
+    lines = out.stdout.strip().splitlines()
+    k, m, n = 8, 2, 512
+    data = [np.array([(j * 53 + i * 7) & 0xFF for i in range(n)], np.uint8)
+            for j in range(k)]
+    exp = oracle.rs_encode(k, m, data, n)
+    got = np.frombuffer(bytes.fromhex(lines[0]), np.uint8).reshape(m, n)
+    for l in range(m):
+        assert np.array_equal(got[l], exp[l]), f"parity {l}"
+    assert lines[1] == "1 1"   # both erased parts recovered bit-exactly
+
+
 def test_reference_headers_link_against_liblizec(tmp_path):
     src = tmp_path / "probe.cc"
     src.write_text(PROBE)
