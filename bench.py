@@ -201,20 +201,28 @@ def main():
                                 4 * (S * STRIPE_BYTES // 65536))
         metric = "GiB/s CRC32 per-64KiB-block"
     elif args.op == "encode_crc":
-        # config 4: encode, then CRC-verify every 64 KiB block of every
-        # part (data re-read + parity read), as the replicator does before
-        # hdd_write (chunk_replicator.cc:189, hddspacemgr.cc:1918)
+        # config 4: encode + CRC-verify every 64 KiB block of every part,
+        # as the replicator does before hdd_write
+        # (chunk_replicator.cc:189, hddspacemgr.cc:1918).  Pipelined: the
+        # data CRC does not depend on the encode, so it runs on a second
+        # stream concurrent with the encode; the parity CRC follows the
+        # encode on the main stream.
         dflat = data.reshape(-1)
         pflat = parity.reshape(-1)
         dcrcs = torch.empty(dflat.numel() // 65536, dtype=torch.int32,
                             device="cuda")
         pcrcs = torch.empty(pflat.numel() // 65536, dtype=torch.int32,
                             device="cuda")
+        crc_stream = torch.cuda.Stream()
 
         def step():
+            main = torch.cuda.current_stream()
+            crc_stream.wait_stream(main)
+            with torch.cuda.stream(crc_stream):
+                lcrc.crc32_blocks(dflat, 65536, out=dcrcs)
             rs.encode_batch(data, parity)
-            lcrc.crc32_blocks(dflat, 65536, out=dcrcs)
             lcrc.crc32_blocks(pflat, 65536, out=pcrcs)
+            main.wait_stream(crc_stream)
         alg_bytes_per_launch = S * STRIPE_BYTES * (K + M) // K  # encode leg
         metric = "GiB/s EC encode+CRC32 pipeline, ec(16,4) 64MiB stripes"
     else:  # mixed — config 5
