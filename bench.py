@@ -214,8 +214,14 @@ def main():
         pcrcs = torch.empty(pflat.numel() // 65536, dtype=torch.int32,
                             device="cuda")
         crc_stream = torch.cuda.Stream()
+        overlap = os.environ.get("LIZEC_BENCH_NO_OVERLAP", "0") != "1"
 
         def step():
+            if not overlap:
+                rs.encode_batch(data, parity)
+                lcrc.crc32_blocks(dflat, 65536, out=dcrcs)
+                lcrc.crc32_blocks(pflat, 65536, out=pcrcs)
+                return
             main = torch.cuda.current_stream()
             crc_stream.wait_stream(main)
             with torch.cuda.stream(crc_stream):
