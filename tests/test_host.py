@@ -39,6 +39,22 @@ def test_exports_complete():
         assert getattr(lib, sym, None) is not None, sym
 
 
+def test_gpu_apis_fail_loudly_without_gpu():
+    """The product path has no CPU fallback: on a GPU-less host, engine
+    creation must raise (LIZEC_ENOGPU), never silently degrade."""
+    import torch
+    if torch.cuda.is_available():
+        pytest.skip("GPU present; the loud-failure path needs a CPU-only host")
+    import lizardfs_amd.lib as _l
+    h = ctypes.c_void_p()
+    rc = _l.lib().lizec_engine_create(ctypes.byref(h), 0)
+    assert rc == -3  # LIZEC_ENOGPU
+    with pytest.raises(L.LizecError, match="ENOGPU"):
+        _l.check(rc, "engine")
+    with pytest.raises(L.LizecError, match="ENOGPU"):
+        _l.engine(0)
+
+
 def test_mangled_crc_aliases_present():
     """The reference's C++-mangled mycrc32 symbols (crc.h:25-31) resolve."""
     lib = L.lib()
