@@ -131,6 +131,9 @@ def main():
     ap.add_argument("--stripes", type=int, default=0,
                     help="stripes per rank (weak scaling); 0 = op default")
     ap.add_argument("--op", choices=tuple(OPS), default="encode")
+    ap.add_argument("--backend", choices=("nccl", "gloo"), default="nccl",
+                    help="gloo allows exercising the multi-rank path on a "
+                         "single GPU (coordination on CPU; compute on GPU)")
     ap.add_argument("--skip-cpu-baseline", action="store_true")
     args = ap.parse_args()
 
@@ -154,9 +157,10 @@ def main():
         log("bench.py requires an MI355X (no CPU fallback on the product path)")
         sys.exit(2)
 
-    torch.cuda.set_device(local_rank)
+    dev_idx = local_rank % torch.cuda.device_count()
+    torch.cuda.set_device(dev_idx)
     if distributed:
-        torch.distributed.init_process_group("nccl")
+        torch.distributed.init_process_group(args.backend)
 
     from lizardfs_amd.ec import ReedSolomon
     from lizardfs_amd import crc as lcrc
@@ -164,13 +168,13 @@ def main():
     workload = (f"ec({K},{M}) {args.op}, {S}x64MiB synthetic "
                 f"stripes/GPU, device-resident")
     log(f"[rank {rank}] generating {S} stripes "
-        f"({S * STRIPE_BYTES / (1 << 30):.0f} GiB data) on cuda:{local_rank}")
+        f"({S * STRIPE_BYTES / (1 << 30):.0f} GiB data) on cuda:{dev_idx}")
     g = torch.Generator(device="cuda").manual_seed(42 + rank)
     data = torch.randint(0, 256, (S, K, PART_LEN), dtype=torch.uint8,
                          device="cuda", generator=g)
     parity = torch.empty((S, M, PART_LEN), dtype=torch.uint8, device="cuda")
 
-    rs = ReedSolomon(K, M, device=local_rank)
+    rs = ReedSolomon(K, M, device=dev_idx)
 
     if args.op == "encode":
         def step():
@@ -283,7 +287,8 @@ def main():
     elapsed = time.perf_counter() - t0
 
     if distributed:
-        t = torch.tensor([elapsed], device="cuda", dtype=torch.float64)
+        dev = "cuda" if args.backend == "nccl" else "cpu"
+        t = torch.tensor([elapsed], device=dev, dtype=torch.float64)
         torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
         elapsed = float(t.item())
 
