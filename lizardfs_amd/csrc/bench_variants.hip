@@ -118,6 +118,7 @@ int main(int argc, char **argv) {
 	    {"D1_CH4_swz_nt ", launch_var<1, 4, true, true>, 4, 1, 262144},
 	    {"D2_CH4_swz_nt ", launch_var<2, 4, true, true>, 4, 2, 262144},
 	    {"D2_CH4_swz_nt_ntld", launch_var<2, 4, true, true, true>, 4, 2, 262144},
+	    {"D2_CH4_ntld_gexact", launch_var<2, 4, true, true, true>, 4, 2, 1048576},
 	    {"D2_CH4_ntld_tp", launch_var<2, 4, true, true, true, true>, 4, 2, 262144},
 	    {"D2_CH6_swz_nt ", launch_var<2, 6, true, true>, 6, 2, 262144},
 	    {"D2_CH4_base   ", launch_var<2, 4, false, false>, 4, 2, 262144},
