@@ -378,7 +378,7 @@ extern "C" int lizec_engine_create(lizec_engine **out, int device_id) {
 	    hipMalloc(&e->d_gftbls, 32 * 32 * 32) != hipSuccess ||
 	    hipMalloc(&e->d_ptrs, e->ptrs_cap * sizeof(uint64_t)) != hipSuccess ||
 	    hipMalloc(&e->d_crc_const, kCrcConstWords * 4) != hipSuccess) {
-		delete e;
+		lizec_engine_destroy(e);   /* frees whatever was allocated */
 		return LIZEC_ENOMEM;
 	}
 	uint32_t *host_const = (uint32_t *)malloc(kCrcConstWords * 4);
