@@ -263,12 +263,14 @@ int oracle_rs_make_tables(int k, int m, uint64_t present_mask,
 	uint8_t recover_m[ORACLE_MAXP * ORACLE_MAXK];
 	uint8_t reduced[ORACLE_MAXP * ORACLE_MAXK];
 	int nparts = k + m;
+	uint64_t all_mask = nparts >= 64 ? ~(uint64_t)0
+	                                 : (((uint64_t)1 << nparts) - 1);
 
 	oracle_init();
 	rs_matrix_for(k, m, rs_matrix);
 
 	/* recover() bookkeeping, reed_solomon.h:97-111 */
-	uint64_t erased = ~present_mask & (((uint64_t)1 << nparts) - 1);
+	uint64_t erased = ~present_mask & all_mask;
 	uint64_t non_zero_input = 0;  /* indexed by surviving-part order */
 	int in_count = 0, out_count = 0, in_with_zero_count = 0;
 	int data_part_count = 0, parity_needed = 0;
@@ -331,7 +333,8 @@ int oracle_rs_recover(int k, int m, const uint8_t **input_fragments,
 	uint8_t *in_parts[ORACLE_MAXP];
 	uint8_t *out_parts[ORACLE_MAXP];
 	int nparts = k + m;
-	uint64_t present = ~erased_mask & (((uint64_t)1 << nparts) - 1);
+	uint64_t present = ~erased_mask &
+	    (nparts >= 64 ? ~(uint64_t)0 : (((uint64_t)1 << nparts) - 1));
 	uint64_t nonnull = 0, needed = 0;
 	int in_count = 0, out_count = 0;
 

@@ -109,6 +109,7 @@ def main():
         (22, 4, "random", (0, 21)),   # k>20, m==4 -> Cauchy branch (reed_solomon.h:168)
         (20, 4, "random", (0, 19)),   # m==4, k<=20 -> Vandermonde branch
         (2, 32, "random", tuple(range(2, 34))[:32]),  # max parity width
+        (32, 32, "random", (0, 15, 31, 40, 50, 63)),  # k+m = 64 edge
     ]
     for ci, (k, m, pat, erase) in enumerate(cases):
         if pat == "counter":
