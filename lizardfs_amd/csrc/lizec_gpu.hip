@@ -52,9 +52,11 @@ constexpr int ec_chunks_for(int d) { return d <= 2 ? 4 : 2; }
 /* ------------------------------------------------------------------ */
 
 /* Device-constant block uploaded once per engine:
- *  [0..4*256)      u32: slicing tables T0..T3 (reflected, poly 0xEDB88320)
- *  [1024..1024+26*32) u32: advance matrices M_i = "append 2^i zero BYTES",
- *                      i = 0..25 (supports block_len < 64 MiB)            */
+ *  [0..16*256)  u32: slicing tables T0..T15 (reflected, poly 0xEDB88320;
+ *               the generic kernel uses T0..T3, the fast path T0..T7 by
+ *               default or T0..T15 under LIZEC_CRC_SLICE=16)
+ *  [then]       u32[26][32]: advance matrices M_i = "append 2^i zero
+ *               BYTES", i = 0..25 (supports block_len < 64 MiB)          */
 constexpr int kCrcTabWords = 16 * 256;  /* slicing-by-16 tables T0..T15 */
 constexpr int kCrcMatCount = 26;
 constexpr int kCrcConstWords = kCrcTabWords + kCrcMatCount * 32;
