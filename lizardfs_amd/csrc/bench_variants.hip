@@ -39,15 +39,16 @@ __global__ void fill_kernel(uint8_t *p, size_t n, uint32_t salt) {
 	}
 }
 
-template <int D, int CH, bool SWZ, bool NTST, bool NTLD = false, bool TPIPE = false>
+template <int D, int CH, bool SWZ, bool NTST, bool NTLD = false,
+          bool TPIPE = false, bool QL = false>
 static void launch_var(uint32_t part_len, int srcs, const uint8_t *tbls,
                        const uint64_t *src, const uint64_t *dst, int dests,
                        uint32_t tiles_per_part, uint32_t total_tiles,
                        uint32_t grid_cap, hipStream_t s) {
 	uint32_t grid = total_tiles < grid_cap ? total_tiles : grid_cap;
-	size_t lds = (size_t)D * srcs * 32;
+	size_t lds = (size_t)D * srcs * (QL ? 16 : 32);
 	for (int base = 0; base + D <= dests; base += D)
-		hipLaunchKernelGGL(HIP_KERNEL_NAME(ec_encode_kernel<D, CH, SWZ, NTST, NTLD, TPIPE>),
+		hipLaunchKernelGGL(HIP_KERNEL_NAME(ec_encode_kernel<D, CH, SWZ, NTST, NTLD, TPIPE, QL>),
 		                   dim3(grid), dim3(kThreads), lds, s, part_len, srcs,
 		                   base, tbls, src, dst, dests, tiles_per_part,
 		                   total_tiles);
@@ -78,6 +79,21 @@ int main(int argc, char **argv) {
 	uint8_t *d_tbls;
 	CK(hipMalloc(&d_tbls, 32 * k * m));
 	CK(hipMemcpy(d_tbls, tbls, 32 * k * m, hipMemcpyHostToDevice));
+	/* packed quarter-LUT layout: 16 B per (dest,src) */
+	static uint8_t tbls_q[16 * 32 * 32];
+	for (int i = 0; i < k * m; ++i) {
+		const uint8_t *t = tbls + i * 32;
+		uint8_t *q = tbls_q + i * 16;
+		for (int b = 0; b < 4; ++b) {
+			q[b] = t[b];
+			q[4 + b] = t[4 * b];
+			q[8 + b] = t[16 + b];
+			q[12 + b] = t[16 + 4 * b];
+		}
+	}
+	uint8_t *d_tbls_q;
+	CK(hipMalloc(&d_tbls_q, 16 * k * m));
+	CK(hipMemcpy(d_tbls_q, tbls_q, 16 * k * m, hipMemcpyHostToDevice));
 
 	std::vector<uint64_t> sp(stripes * k), dp(stripes * m);
 	for (int s = 0; s < stripes; ++s) {
@@ -113,12 +129,14 @@ int main(int argc, char **argv) {
 		int ch;
 		int d;
 		uint32_t grid_cap;
+		bool ql = false;
 	};
 	Cfg cfgs[] = {
 	    {"D1_CH4_swz_nt ", launch_var<1, 4, true, true>, 4, 1, 262144},
 	    {"D2_CH4_swz_nt ", launch_var<2, 4, true, true>, 4, 2, 262144},
 	    {"D2_CH4_swz_nt_ntld", launch_var<2, 4, true, true, true>, 4, 2, 262144},
 	    {"D2_CH4_ntld_gexact", launch_var<2, 4, true, true, true>, 4, 2, 1048576},
+	    {"D2_CH4_ql     ", launch_var<2, 4, true, true, true, false, true>, 4, 2, 1048576, true},
 	    {"D2_CH4_ntld_tp", launch_var<2, 4, true, true, true, true>, 4, 2, 262144},
 	    {"D2_CH6_swz_nt ", launch_var<2, 6, true, true>, 6, 2, 262144},
 	    {"D2_CH4_base   ", launch_var<2, 4, false, false>, 4, 2, 262144},
@@ -127,6 +145,8 @@ int main(int argc, char **argv) {
 	    {"D4_CH3_swz_nt ", launch_var<4, 3, true, true>, 3, 4, 262144},
 	    {"D4_CH2_swz_nt ", launch_var<4, 2, true, true>, 2, 4, 262144},
 	    {"D4_CH2_ntld   ", launch_var<4, 2, true, true, true>, 2, 4, 262144},
+	    {"D4_CH2_ql     ", launch_var<4, 2, true, true, true, false, true>, 2, 4, 1048576, true},
+	    {"D4_CH4_ql     ", launch_var<4, 4, true, true, true, false, true>, 4, 4, 1048576, true},
 	    {"D4_CH2_ntld_tp", launch_var<4, 2, true, true, true, true>, 2, 4, 262144},
 	    {"D4_CH4_ntld_tp", launch_var<4, 4, true, true, true, true>, 4, 4, 262144},
 	    {"D4_CH4_ntld   ", launch_var<4, 4, true, true, true>, 4, 4, 262144},
@@ -135,6 +155,7 @@ int main(int argc, char **argv) {
 	    {"D6_CH3_swz_nt ", launch_var<6, 3, true, true>, 3, 6, 262144},
 	    {"D6_CH2_swz_nt ", launch_var<6, 2, true, true>, 2, 6, 262144},
 	    {"D6_CH2_ntld   ", launch_var<6, 2, true, true, true>, 2, 6, 262144},
+	    {"D6_CH2_ql     ", launch_var<6, 2, true, true, true, false, true>, 2, 6, 1048576, true},
 	    {"D6_CH2_ntld_tp", launch_var<6, 2, true, true, true, true>, 2, 6, 262144},
 	    {"D6_CH4_ntld   ", launch_var<6, 4, true, true, true>, 4, 6, 262144},
 	    {"D8_CH2_swz_nt ", launch_var<8, 2, true, true>, 2, 8, 262144},
@@ -153,13 +174,14 @@ int main(int argc, char **argv) {
 		uint32_t tot = tpp * stripes;
 		CK(hipMemset(d_par, 0, 4096));
 		/* warmup x2 */
+		const uint8_t *tb = c.ql ? d_tbls_q : d_tbls;
 		for (int r = 0; r < 2; ++r)
-			c.fn((uint32_t)part_len, k, d_tbls, d_sp, d_dp, m, tpp, tot,
+			c.fn((uint32_t)part_len, k, tb, d_sp, d_dp, m, tpp, tot,
 			     c.grid_cap, 0);
 		CK(hipDeviceSynchronize());
 		CK(hipEventRecord(e0, 0));
 		for (int r = 0; r < reps; ++r)
-			c.fn((uint32_t)part_len, k, d_tbls, d_sp, d_dp, m, tpp, tot,
+			c.fn((uint32_t)part_len, k, tb, d_sp, d_dp, m, tpp, tot,
 			     c.grid_cap, 0);
 		CK(hipEventRecord(e1, 0));
 		CK(hipDeviceSynchronize());

@@ -57,6 +57,47 @@ __device__ __forceinline__ void gf_macc_all(uint4 (&acc)[D][CH], int c,
 	}
 }
 
+/* Quarter-LUT GF multiply: byte n = f0 + 4*f1 + 16*f2 + 64*f3 (2-bit
+ * fields, disjoint bits => XOR), so c*n = c*f0 ^ (4c)*f1 ^ (16c)*f2 ^
+ * (64c)*f3 by GF(2^8) linearity.  Each 4-entry product table packs into
+ * ONE register and one v_perm (sel 0-3 picks src1 bytes), so a dest costs
+ * 4 perms + xor-fold instead of 6 perms + selector math — and the LDS
+ * table is 16 bytes per (src,dest) instead of 32.  Packed layout (from
+ * the ISA-L 32-byte table): [tbl[0..3], tbl[{0,4,8,12}], tbl[16..19],
+ * tbl[16+{0,4,8,12}]]. */
+__device__ __forceinline__ uint32_t gf_macc_q(uint32_t acc, const uint4 &T,
+                                              uint32_t f0, uint32_t f1,
+                                              uint32_t f2, uint32_t f3) {
+	uint32_t p0 = __builtin_amdgcn_perm(T.x, T.x, f0);
+	uint32_t p1 = __builtin_amdgcn_perm(T.y, T.y, f1);
+	uint32_t p2 = __builtin_amdgcn_perm(T.z, T.z, f2);
+	uint32_t p3 = __builtin_amdgcn_perm(T.w, T.w, f3);
+	return acc ^ p0 ^ p1 ^ p2 ^ p3;
+}
+
+template <int D, int CH>
+__device__ __forceinline__ void gf_macc_all_q(uint4 (&acc)[D][CH], int c,
+                                              const uint4 &w,
+                                              const uint4 (&T)[D]) {
+	constexpr uint32_t M = 0x03030303u;
+	uint32_t f[4][4];
+	const uint32_t ws[4] = {w.x, w.y, w.z, w.w};
+#pragma unroll
+	for (int q = 0; q < 4; ++q) {
+		f[q][0] = ws[q] & M;
+		f[q][1] = (ws[q] >> 2) & M;
+		f[q][2] = (ws[q] >> 4) & M;
+		f[q][3] = (ws[q] >> 6) & M;
+	}
+#pragma unroll
+	for (int d = 0; d < D; ++d) {
+		acc[d][c].x = gf_macc_q(acc[d][c].x, T[d], f[0][0], f[0][1], f[0][2], f[0][3]);
+		acc[d][c].y = gf_macc_q(acc[d][c].y, T[d], f[1][0], f[1][1], f[1][2], f[1][3]);
+		acc[d][c].z = gf_macc_q(acc[d][c].z, T[d], f[2][0], f[2][1], f[2][2], f[2][3]);
+		acc[d][c].w = gf_macc_q(acc[d][c].w, T[d], f[3][0], f[3][1], f[3][2], f[3][3]);
+	}
+}
+
 /* Bijective XCD remap (8 XCDs): consecutive hardware block ids round-robin
  * the XCDs; remapped ids give each XCD one contiguous range. */
 __device__ __forceinline__ uint32_t xcd_remap(uint32_t b, uint32_t n) {
@@ -73,7 +114,7 @@ __device__ __forceinline__ uint4 ld_nt(const uint8_t *p) {
 }
 
 template <int D, int CH, bool SWZ, bool NTST, bool NTLD = false,
-          bool TPIPE = false>
+          bool TPIPE = false, bool QL = false>
 __global__ __launch_bounds__(kThreads) void ec_encode_kernel(
     uint32_t part_len, int srcs, int dest_base,
     const uint8_t *__restrict__ gftbls_dev,  /* 32*srcs*dests_total */
@@ -84,10 +125,12 @@ __global__ __launch_bounds__(kThreads) void ec_encode_kernel(
 	extern __shared__ __attribute__((aligned(16))) uint8_t smem[];
 	const uint32_t tid = threadIdx.x;
 
-	/* Stage this pass's table rows: [D][srcs][32] bytes. */
+	/* Stage this pass's table rows: [D][srcs][32] bytes (16 when QL —
+	 * gftbls_dev then holds the packed quarter-LUT layout). */
 	{
-		const uint8_t *src_tbl = gftbls_dev + (size_t)dest_base * srcs * 32;
-		int nbytes = D * srcs * 32;
+		const int tw = QL ? 16 : 32;
+		const uint8_t *src_tbl = gftbls_dev + (size_t)dest_base * srcs * tw;
+		int nbytes = D * srcs * tw;
 		for (int i = tid * 16; i < nbytes; i += kThreads * 16)
 			*(uint4 *)(smem + i) = *(const uint4 *)(src_tbl + i);
 	}
@@ -115,13 +158,19 @@ __global__ __launch_bounds__(kThreads) void ec_encode_kernel(
 			 * current source is accumulated (the per-source LDS
 			 * broadcast + lgkmcnt wait otherwise serializes large k) */
 			uint4 w[CH], wn[CH];
-			uint4 L[TPIPE ? 2 : 1][D], H[TPIPE ? 2 : 1][D];
+			uint4 L[TPIPE ? 2 : 1][D], H[QL ? 1 : (TPIPE ? 2 : 1)][D];
 			auto tbl_read = [&](int j, uint4 (&Lb)[D], uint4 (&Hb)[D]) {
 #pragma unroll
 				for (int d = 0; d < D; ++d) {
-					const uint8_t *tb = smem + ((size_t)d * srcs + j) * 32;
-					Lb[d] = *(const uint4 *)tb;
-					Hb[d] = *(const uint4 *)(tb + 16);
+					if (QL) {
+						Lb[d] = *(const uint4 *)(smem +
+						                         ((size_t)d * srcs + j) * 16);
+					} else {
+						const uint8_t *tb =
+						    smem + ((size_t)d * srcs + j) * 32;
+						Lb[d] = *(const uint4 *)tb;
+						Hb[d] = *(const uint4 *)(tb + 16);
+					}
 				}
 			};
 			{
@@ -147,8 +196,12 @@ __global__ __launch_bounds__(kThreads) void ec_encode_kernel(
 				}
 				if (!TPIPE) tbl_read(j, Lc, Hc);
 #pragma unroll
-				for (int c = 0; c < CH; ++c)
-					gf_macc_all<D, CH>(acc, c, w[c], Lc, Hc);
+				for (int c = 0; c < CH; ++c) {
+					if (QL)
+						gf_macc_all_q<D, CH>(acc, c, w[c], Lc);
+					else
+						gf_macc_all<D, CH>(acc, c, w[c], Lc, Hc);
+				}
 #pragma unroll
 				for (int c = 0; c < CH; ++c) w[c] = wn[c];
 			};
