@@ -236,16 +236,25 @@ __global__ __launch_bounds__(kThreads) void ec_encode_kernel(
 				uint4 L[D], H[D];
 #pragma unroll
 				for (int d = 0; d < D; ++d) {
-					const uint8_t *tb = smem + ((size_t)d * srcs + j) * 32;
-					L[d] = *(const uint4 *)tb;
-					H[d] = *(const uint4 *)(tb + 16);
+					if (QL) {
+						L[d] = *(const uint4 *)(smem +
+						                        ((size_t)d * srcs + j) * 16);
+					} else {
+						const uint8_t *tb =
+						    smem + ((size_t)d * srcs + j) * 32;
+						L[d] = *(const uint4 *)tb;
+						H[d] = *(const uint4 *)(tb + 16);
+					}
 				}
 #pragma unroll
 				for (int c = 0; c < CH; ++c) {
 					uint32_t off = base + c * kChunkBytes;
 					if (off < part_len) {
 						uint4 w = *(const uint4 *)(sp + off);
-						gf_macc_all<D, CH>(acc, c, w, L, H);
+						if (QL)
+							gf_macc_all_q<D, CH>(acc, c, w, L);
+						else
+							gf_macc_all<D, CH>(acc, c, w, L, H);
 					}
 				}
 			}
