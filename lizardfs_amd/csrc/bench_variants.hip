@@ -156,6 +156,9 @@ int main(int argc, char **argv) {
 	    {"D6_CH2_swz_nt ", launch_var<6, 2, true, true>, 2, 6, 262144},
 	    {"D6_CH2_ntld   ", launch_var<6, 2, true, true, true>, 2, 6, 262144},
 	    {"D6_CH2_ql     ", launch_var<6, 2, true, true, true, false, true>, 2, 6, 1048576, true},
+	    {"D6_CH4_ql     ", launch_var<6, 4, true, true, true, false, true>, 4, 6, 1048576, true},
+	    {"D8_CH2_ql     ", launch_var<8, 2, true, true, true, false, true>, 2, 8, 1048576, true},
+	    {"D8_CH4_ql     ", launch_var<8, 4, true, true, true, false, true>, 4, 8, 1048576, true},
 	    {"D6_CH2_ntld_tp", launch_var<6, 2, true, true, true, true>, 2, 6, 262144},
 	    {"D6_CH4_ntld   ", launch_var<6, 4, true, true, true>, 4, 6, 262144},
 	    {"D8_CH2_swz_nt ", launch_var<8, 2, true, true>, 2, 8, 262144},
