@@ -1,8 +1,9 @@
 /* bench_variants.hip — A/B harness for the EC encode kernel variants on a
  * real MI355X.  Standalone executable (not part of liblizec.so): allocates
- * device-resident ec(8,2)-shaped batches, runs each template configuration,
- * checks a sample of the output against the host scalar path, and prints a
- * GB/s table (traffic = (k+m)/k bytes per data byte).
+ * device-resident ec(k,m) batches (k, m, stripes, reps from argv), runs
+ * each template configuration whose D divides m, checks a sample of the
+ * output against the host scalar path, and prints a GB/s table
+ * (traffic-based: (k+m)/k bytes per data byte).
  *
  * Build: make bench_variants  ; run on the GPU box only.
  */
