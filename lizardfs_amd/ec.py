@@ -132,6 +132,9 @@ class ReedSolomon:
                                  device=data.device)
         else:
             self._check_part(parity, "parity")
+            if tuple(parity.shape) != (S, self.m, plen):
+                raise ValueError(f"parity must be [S={S}, m={self.m}, "
+                                 f"L={plen}], got {tuple(parity.shape)}")
 
         dmask = (1 << self.k) - 1
         pmask = ((1 << self.m) - 1) << self.k
@@ -181,6 +184,9 @@ class ReedSolomon:
             present |= 1 << i
             if fragments[i] is not None:
                 s, l, _ = self._frag_geom(fragments[i], f"fragment {i}")
+                if S is not None and (s, l) != (S, plen):
+                    raise ValueError(
+                        f"fragment {i} is [{s}, {l}], expected [{S}, {plen}]")
                 nonnull |= 1 << i
                 S, plen = s, l
                 dev = fragments[i].device
