@@ -121,6 +121,9 @@ __global__ __launch_bounds__(kThreads) void ec_encode_kernel(
     const uint64_t *__restrict__ src_ptrs,   /* [stripes][srcs]  */
     const uint64_t *__restrict__ dst_ptrs,   /* [stripes][dests_total] */
     int dests_total, uint32_t tiles_per_part, uint32_t total_tiles) {
+	/* QL keeps its packed table in L only; H is sized 1 then, so the
+	 * TPIPE double-buffer path (H[1]) must not be combined with it. */
+	static_assert(!(QL && TPIPE), "QL and TPIPE are mutually exclusive");
 	constexpr uint32_t kTile = kChunkBytes * CH;
 	extern __shared__ __attribute__((aligned(16))) uint8_t smem[];
 	const uint32_t tid = threadIdx.x;

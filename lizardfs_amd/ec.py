@@ -205,6 +205,9 @@ class ReedSolomon:
             outs = out   # caller-provided stable buffers -> plan is cached
             for i in want:
                 self._check_part(outs[i], f"out[{i}]")
+                if tuple(outs[i].shape) != (S, plen):
+                    raise ValueError(f"out[{i}] must be [S={S}, L={plen}], "
+                                     f"got {tuple(outs[i].shape)}")
         else:
             outs = {i: torch.empty((S, plen), dtype=torch.uint8, device=dev)
                     for i in sorted(want)}

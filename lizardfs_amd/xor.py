@@ -48,6 +48,10 @@ class XorSlice:
     def _xor_reduce(self, parts, out):
         """out[s] = XOR of parts[j][s] over j (device, via the EC kernel)."""
         S, plen = parts[0].shape[0], parts[0].shape[-1]
+        if (out.dtype != torch.uint8 or not out.is_cuda
+                or not out.is_contiguous() or tuple(out.shape) != (S, plen)):
+            raise ValueError(f"out must be a contiguous CUDA uint8 "
+                             f"[S={S}, L={plen}] tensor")
         srcs = len(parts)
         tbl = self._ones_tables(srcs)
         rows = np.arange(S, dtype=np.uint64)
@@ -55,6 +59,9 @@ class XorSlice:
         for j, t in enumerate(parts):
             if t.dtype != torch.uint8 or not t.is_cuda or t.stride(-1) != 1:
                 raise ValueError("parts must be CUDA uint8 with contiguous rows")
+            if tuple(t.shape) != (S, plen):
+                raise ValueError(f"part {j} is {tuple(t.shape)}, "
+                                 f"expected [{S}, {plen}]")
             src[:, j] = t.data_ptr() + rows * np.uint64(t.stride(0))
         dst = (out.data_ptr() + rows * np.uint64(plen)).astype(np.uint64)
         stream = torch.cuda.current_stream(self.device).cuda_stream
