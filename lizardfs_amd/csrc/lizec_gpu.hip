@@ -25,6 +25,8 @@
 #include <cstdio>
 #include <cstring>
 #include <cstdlib>
+#include <mutex>
+#include <unordered_map>
 
 #include "../../include/lizec.h"
 
@@ -35,6 +37,7 @@
 	} while (0)
 
 #include "ec_kernel.h"
+#include "crc_fold.h"
 
 /* Product kernel configuration (chosen by the variant A/B harness,
  * bench_variants.hip; numbers in profiles/ROUND1.md):
@@ -72,28 +75,6 @@ static void pack_quarter_lut(const uint8_t *t, uint8_t *q) {
 constexpr int kCrcTabWords = 16 * 256;  /* slicing-by-16 tables T0..T15 */
 constexpr int kCrcMatCount = 26;
 constexpr int kCrcConstWords = kCrcTabWords + kCrcMatCount * 32;
-
-/* Apply advance-by-len2-zero-bytes to crc (mycrc32_combine semantics,
- * crc.cc:207-224) using the LDS matrix bank. */
-__device__ __forceinline__ uint32_t crc_advance(uint32_t crc, uint32_t len2,
-                                                const uint32_t *mats) {
-	int i = 0;
-	while (len2) {
-		if (len2 & 1) {
-			const uint32_t *M = mats + i * 32;
-			uint32_t r = 0, v = crc;
-#pragma unroll
-			for (int j = 0; j < 32; ++j) {
-				r ^= (v & 1) ? M[j] : 0u;
-				v >>= 1;
-			}
-			crc = r;
-		}
-		len2 >>= 1;
-		++i;
-	}
-	return crc;
-}
 
 /* One wave per block: lane l owns segment [l*seg, l*seg+seg) of the block
  * (seg = block_len/64, multiple of 16 enforced by the host).  Lane 0 seeds;
@@ -271,6 +252,31 @@ __global__ __launch_bounds__(kThreads) void crc32_blocks_kernel_multi(
 }
 
 
+/* LDS-free folding CRC kernel (crc_fold.h): the hot loop is pure VALU;
+ * LDS holds only the epilogue byte table T0 and the combine matrices. */
+constexpr int kCrcFoldLdsWords = 256 + kCrcMatCount * 32;
+
+template <int C, int NACC, bool AL16 = true>
+__global__ __launch_bounds__(kThreads) void crc32_blocks_kernel_fold(
+    const uint8_t *__restrict__ buf, uint32_t block_len, uint64_t nblocks,
+    uint32_t seed, const uint32_t *__restrict__ crc_const,
+    uint32_t *__restrict__ out) {
+	__shared__ __attribute__((aligned(16))) uint32_t stabs[kCrcFoldLdsWords];
+	for (int i = threadIdx.x; i < 256; i += kThreads)
+		stabs[i] = crc_const[i];
+	for (int i = threadIdx.x; i < kCrcMatCount * 32; i += kThreads)
+		stabs[256 + i] = crc_const[kCrcTabWords + i];
+	__syncthreads();
+	const int wave = threadIdx.x >> 6;
+	const int lane = threadIdx.x & 63;
+	for (uint64_t blk = (uint64_t)blockIdx.x * 4 + wave; blk < nblocks;
+	     blk += (uint64_t)gridDim.x * 4) {
+		uint32_t crc = crc_block_wave_fold<C, NACC, AL16>(
+		    buf + blk * block_len, block_len, seed, stabs, stabs + 256, lane);
+		if (lane == 0) out[blk] = crc;
+	}
+}
+
 /* Batched chunk scrub — hdd_int_test semantics (hddspacemgr.cc:2148-2212):
  * for every 64 KiB block of every MooseFS-format chunk-part image, compare
  * mycrc32(0, block, MFSBLOCKSIZE) against the stored CRC array entry
@@ -284,12 +290,14 @@ __global__ __launch_bounds__(kThreads) void scrub_chunks_kernel(
     const uint32_t *__restrict__ block_counts, uint32_t nchunks,
     uint32_t max_blocks, uint32_t block_stride, uint32_t crc_stride,
     const uint32_t *__restrict__ crc_const, int32_t *__restrict__ status) {
-	__shared__ __attribute__((aligned(16))) uint32_t stabs[kCrcConstWords];
-	for (int i = threadIdx.x; i < kCrcConstWords; i += kThreads)
+	__shared__ __attribute__((aligned(16))) uint32_t stabs[kCrcFoldLdsWords];
+	for (int i = threadIdx.x; i < 256; i += kThreads)
 		stabs[i] = crc_const[i];
+	for (int i = threadIdx.x; i < kCrcMatCount * 32; i += kThreads)
+		stabs[256 + i] = crc_const[kCrcTabWords + i];
 	__syncthreads();
-	const uint32_t *T = stabs;
-	const uint32_t *mats = stabs + kCrcTabWords;
+	const uint32_t *T0 = stabs;
+	const uint32_t *mats = stabs + 256;
 	const int wave = threadIdx.x >> 6;
 	const int lane = threadIdx.x & 63;
 	const uint64_t total = (uint64_t)nchunks * max_blocks;
@@ -299,8 +307,15 @@ __global__ __launch_bounds__(kThreads) void scrub_chunks_kernel(
 		uint32_t b = (uint32_t)(flat - (uint64_t)c * max_blocks);
 		if (b >= block_counts[c]) continue;
 		const uint8_t *img = (const uint8_t *)chunk_dptrs[c];
-		uint32_t crc = crc_block_wave<2, 8, 8>(
-		    img + data_offs[c] + b * block_stride, 65536u, 0u, T, mats, lane);
+		const uint8_t *blockp = img + data_offs[c] + b * block_stride;
+		/* INTERLEAVED-format blocks sit at 4 mod 16 — use the dword-load
+		 * instantiation there (uint4 loads would be misaligned UB) */
+		uint32_t crc =
+		    (((uintptr_t)blockp & 15) == 0)
+		        ? crc_block_wave_fold<2, 2, true>(blockp, 65536u, 0u, T0,
+		                                          mats, lane)
+		        : crc_block_wave_fold<2, 2, false>(blockp, 65536u, 0u, T0,
+		                                           mats, lane);
 		if (lane == 0) {
 			const uint8_t *p = img + crc_offs[c] + crc_stride * b;
 			uint32_t stored = ((uint32_t)p[0] << 24) | ((uint32_t)p[1] << 16) |
@@ -314,14 +329,87 @@ __global__ __launch_bounds__(kThreads) void scrub_chunks_kernel(
 /* Engine + plans                                                     */
 /* ------------------------------------------------------------------ */
 
+/* Per-stream call scratch.  The batch entry points upload tables/pointer
+ * arrays into device scratch; keying the scratch by stream makes
+ * concurrent calls on different streams of one engine race-free (the
+ * chunkserver's bgjobs threading model is exactly multi-threaded,
+ * network_main_thread.cc:230-234).  Within one stream, device scratch
+ * reuse is stream-ordered; the pinned host staging buffer is guarded by
+ * an event recorded after the H2D copies are enqueued. */
+struct lizec_stream_ctx {
+	uint8_t *d_gftbls = nullptr;    /* 16*32*32 (packed quarter-LUT) */
+	uint64_t *d_ptrs = nullptr;     /* grows */
+	size_t ptrs_cap = 0;            /* in elements */
+	uint8_t *h_staging = nullptr;   /* pinned: tables + pointer arrays */
+	size_t h_cap = 0;               /* staging bytes */
+	hipEvent_t uploaded = nullptr;  /* staging consumed up to here */
+	bool ev_recorded = false;
+};
+
 struct lizec_engine {
 	int device;
 	hipStream_t stream;        /* default stream for calls passing NULL */
-	uint8_t *d_gftbls;         /* 32*32*32 max (per-call scratch) */
-	uint64_t *d_ptrs;          /* per-call scratch, grows */
-	size_t ptrs_cap;           /* in elements */
 	uint32_t *d_crc_const;     /* kCrcConstWords */
+	std::mutex mu;             /* guards ctxs */
+	std::unordered_map<void *, lizec_stream_ctx> ctxs;
 };
+
+static void ctx_free(lizec_stream_ctx &c) {
+	(void)hipFree(c.d_gftbls);
+	(void)hipFree(c.d_ptrs);
+	(void)hipHostFree(c.h_staging);
+	if (c.uploaded) (void)hipEventDestroy(c.uploaded);
+	c = lizec_stream_ctx();
+}
+
+/* Get (or create) the scratch for `s`, sized for `ptrs` pointer slots and
+ * `staging` staging bytes; waits out any still-pending staging use.
+ * Thread contract: concurrent calls must use distinct streams (each
+ * stream's scratch is touched by one call at a time — the map lookup is
+ * the only globally locked step, so streams never stall each other). */
+static int ctx_acquire(lizec_engine *e, hipStream_t s, size_t ptrs,
+                       size_t staging, lizec_stream_ctx **out) {
+	lizec_stream_ctx *cp;
+	{
+		std::lock_guard<std::mutex> lk(e->mu);
+		cp = &e->ctxs[(void *)s];   /* element refs survive rehash */
+	}
+	lizec_stream_ctx &c = *cp;
+	if (!c.uploaded) {
+		if (hipEventCreateWithFlags(&c.uploaded, hipEventDisableTiming) !=
+		    hipSuccess)
+			return LIZEC_EHIP;
+	}
+	if (c.ev_recorded) {
+		/* previous call on this stream may still be reading h_staging (and
+		 * a grow below would free device scratch it reads) — wait it out */
+		LIZEC_CHECK(hipEventSynchronize(c.uploaded));
+		c.ev_recorded = false;
+	}
+	if (!c.d_gftbls)
+		LIZEC_CHECK(hipMalloc(&c.d_gftbls, (size_t)16 * 32 * 32));
+	if (c.ptrs_cap < ptrs) {
+		size_t cap = c.ptrs_cap ? c.ptrs_cap : (1 << 16);
+		while (cap < ptrs) cap *= 2;
+		(void)hipFree(c.d_ptrs);
+		c.d_ptrs = nullptr;
+		c.ptrs_cap = 0;
+		LIZEC_CHECK(hipMalloc(&c.d_ptrs, cap * sizeof(uint64_t)));
+		c.ptrs_cap = cap;
+	}
+	size_t want = staging + 16 * 32 * 32;
+	if (c.h_cap < want) {
+		size_t cap = c.h_cap ? c.h_cap : (1 << 20);
+		while (cap < want) cap *= 2;
+		(void)hipHostFree(c.h_staging);
+		c.h_staging = nullptr;
+		c.h_cap = 0;
+		LIZEC_CHECK(hipHostMalloc(&c.h_staging, cap));
+		c.h_cap = cap;
+	}
+	*out = &c;
+	return LIZEC_OK;
+}
 
 /* A plan = a prepared batch: device-resident tables + pointer arrays.
  * Mirrors the reference's cached-matrix + read-plan structure
@@ -387,10 +475,7 @@ extern "C" int lizec_engine_create(lizec_engine **out, int device_id) {
 	LIZEC_CHECK(hipSetDevice(device_id));
 	lizec_engine *e = new lizec_engine();
 	e->device = device_id;
-	e->ptrs_cap = 1 << 16;
 	if (hipStreamCreate(&e->stream) != hipSuccess ||
-	    hipMalloc(&e->d_gftbls, 32 * 32 * 32) != hipSuccess ||
-	    hipMalloc(&e->d_ptrs, e->ptrs_cap * sizeof(uint64_t)) != hipSuccess ||
 	    hipMalloc(&e->d_crc_const, kCrcConstWords * 4) != hipSuccess) {
 		lizec_engine_destroy(e);   /* frees whatever was allocated */
 		return LIZEC_ENOMEM;
@@ -410,8 +495,7 @@ extern "C" int lizec_engine_create(lizec_engine **out, int device_id) {
 
 extern "C" void lizec_engine_destroy(lizec_engine *e) {
 	if (!e) return;
-	(void)hipFree(e->d_gftbls);
-	(void)hipFree(e->d_ptrs);
+	for (auto &kv : e->ctxs) ctx_free(kv.second);
 	(void)hipFree(e->d_crc_const);
 	(void)hipStreamDestroy(e->stream);
 	delete e;
@@ -420,18 +504,6 @@ extern "C" void lizec_engine_destroy(lizec_engine *e) {
 extern "C" int lizec_engine_sync(lizec_engine *e) {
 	LIZEC_CHECK(hipSetDevice(e->device));
 	LIZEC_CHECK(hipStreamSynchronize(e->stream));
-	return LIZEC_OK;
-}
-
-static int ensure_ptrs(lizec_engine *e, size_t n) {
-	if (n <= e->ptrs_cap) return LIZEC_OK;
-	size_t cap = e->ptrs_cap;
-	while (cap < n) cap *= 2;
-	uint64_t *p = nullptr;
-	LIZEC_CHECK(hipMalloc(&p, cap * sizeof(uint64_t)));
-	LIZEC_CHECK(hipFree(e->d_ptrs));
-	e->d_ptrs = p;
-	e->ptrs_cap = cap;
 	return LIZEC_OK;
 }
 
@@ -519,23 +591,27 @@ extern "C" int lizec_ec_encode_batch(lizec_engine *e, uint64_t part_len,
 
 	size_t nsrc = (size_t)num_stripes * srcs;
 	size_t ndst = (size_t)num_stripes * dests;
-	r = ensure_ptrs(e, nsrc + ndst);
+	lizec_stream_ctx *c;
+	r = ctx_acquire(e, s, nsrc + ndst, (nsrc + ndst) * 8, &c);
 	if (r != LIZEC_OK) return r;
-	uint64_t *d_src = e->d_ptrs;
-	uint64_t *d_dst = e->d_ptrs + nsrc;
-	/* pageable-source async copy is consumed before return, so the stack
-	 * buffer is safe; the write stays ordered on stream s */
-	uint8_t packed[16 * 32 * 32];
+	uint64_t *d_src = c->d_ptrs;
+	uint64_t *d_dst = c->d_ptrs + nsrc;
+	/* stage tables + pointer arrays through this stream's pinned buffer
+	 * (ctx_acquire waited until any previous copies from it finished) */
+	uint8_t *tstage = c->h_staging;
+	uint64_t *pstage = (uint64_t *)(c->h_staging + 16 * 32 * 32);
 	for (int i = 0; i < srcs * dests; ++i)
-		pack_quarter_lut(gftbls + (size_t)i * 32, packed + (size_t)i * 16);
-	LIZEC_CHECK(hipMemcpyAsync(e->d_gftbls, packed, (size_t)16 * srcs * dests,
+		pack_quarter_lut(gftbls + (size_t)i * 32, tstage + (size_t)i * 16);
+	memcpy(pstage, src_dptrs, nsrc * 8);
+	memcpy(pstage + nsrc, dst_dptrs, ndst * 8);
+	LIZEC_CHECK(hipMemcpyAsync(c->d_gftbls, tstage, (size_t)16 * srcs * dests,
 	                           hipMemcpyHostToDevice, s));
-	LIZEC_CHECK(hipMemcpyAsync(d_src, src_dptrs, nsrc * 8,
+	LIZEC_CHECK(hipMemcpyAsync(d_src, pstage, (nsrc + ndst) * 8,
 	                           hipMemcpyHostToDevice, s));
-	LIZEC_CHECK(hipMemcpyAsync(d_dst, dst_dptrs, ndst * 8,
-	                           hipMemcpyHostToDevice, s));
+	LIZEC_CHECK(hipEventRecord(c->uploaded, s));
+	c->ev_recorded = true;
 
-	return run_batch(part_len, srcs, dests, e->d_gftbls, d_src, d_dst,
+	return run_batch(part_len, srcs, dests, c->d_gftbls, d_src, d_dst,
 	                 (uint32_t)num_stripes, s);
 }
 
@@ -609,8 +685,58 @@ extern "C" int lizec_crc32_batch(lizec_engine *e, const void *dev_buf,
 	uint32_t grid = (uint32_t)(groups < 131072 ? groups : 131072);
 	const char *ch = getenv("LIZEC_CRC_CHAINS");   /* A/B hooks */
 	const char *sl = getenv("LIZEC_CRC_SLICE");
+	const char *im = getenv("LIZEC_CRC_IMPL");
+	const char *fn = getenv("LIZEC_CRC_FOLD_NACC");
 	int chains = ch ? atoi(ch) : 2;   /* C=4 measured worse: half-line bursts re-thrash L1 (profiles) */
 	int slice = sl ? atoi(sl) : 8;   /* slice-16 measured -16% within-box (profiles) */
+	bool fold = !(im && strcmp(im, "table") == 0);
+	int nacc = fn ? atoi(fn) : 2;
+	bool al16 = (((uintptr_t)dev_buf | block_len) & 15) == 0;
+	/* carry-less-folding path (default): block must split into C spans of
+	 * whole 64-lane x BV*16-byte bursts */
+	if (fold && block_len % (2 * 64 * 16 * 8) == 0 && chains == 2) {
+		const uint8_t *b = (const uint8_t *)dev_buf;
+#define LIZEC_LAUNCH_FOLD(C, N, A)                                          \
+	hipLaunchKernelGGL(HIP_KERNEL_NAME(crc32_blocks_kernel_fold<C, N, A>),  \
+	                   dim3(grid), dim3(kThreads), 0, s, b, block_len,      \
+	                   nblocks, seed, e->d_crc_const, dev_crcs_out)
+		if (al16) {
+			if (nacc == 1) LIZEC_LAUNCH_FOLD(2, 1, true);
+			else if (nacc == 4) LIZEC_LAUNCH_FOLD(2, 4, true);
+			else LIZEC_LAUNCH_FOLD(2, 2, true);
+		} else {
+			LIZEC_LAUNCH_FOLD(2, 2, false);
+		}
+#undef LIZEC_LAUNCH_FOLD
+		LIZEC_CHECK(hipGetLastError());
+		return LIZEC_OK;
+	}
+	if (fold && block_len % (4 * 64 * 16 * 4) == 0 && chains == 4) {
+		const uint8_t *b = (const uint8_t *)dev_buf;
+		if (al16)
+			hipLaunchKernelGGL(HIP_KERNEL_NAME(crc32_blocks_kernel_fold<4, 1, true>),
+			                   dim3(grid), dim3(kThreads), 0, s, b, block_len,
+			                   nblocks, seed, e->d_crc_const, dev_crcs_out);
+		else
+			hipLaunchKernelGGL(HIP_KERNEL_NAME(crc32_blocks_kernel_fold<4, 1, false>),
+			                   dim3(grid), dim3(kThreads), 0, s, b, block_len,
+			                   nblocks, seed, e->d_crc_const, dev_crcs_out);
+		LIZEC_CHECK(hipGetLastError());
+		return LIZEC_OK;
+	}
+	if (fold && block_len % (64 * 16 * 8) == 0 && chains == 1) {
+		const uint8_t *b = (const uint8_t *)dev_buf;
+		if (al16)
+			hipLaunchKernelGGL(HIP_KERNEL_NAME(crc32_blocks_kernel_fold<1, 2, true>),
+			                   dim3(grid), dim3(kThreads), 0, s, b, block_len,
+			                   nblocks, seed, e->d_crc_const, dev_crcs_out);
+		else
+			hipLaunchKernelGGL(HIP_KERNEL_NAME(crc32_blocks_kernel_fold<1, 2, false>),
+			                   dim3(grid), dim3(kThreads), 0, s, b, block_len,
+			                   nblocks, seed, e->d_crc_const, dev_crcs_out);
+		LIZEC_CHECK(hipGetLastError());
+		return LIZEC_OK;
+	}
 	if (block_len % 32768 == 0 && chains >= 4)
 		hipLaunchKernelGGL(HIP_KERNEL_NAME(crc32_blocks_kernel_multi<4, 8>),
 		                   dim3(grid), dim3(kThreads), 0, s,
@@ -643,23 +769,27 @@ extern "C" int lizec_scrub_batch_strided(
 		return LIZEC_EINVAL;
 	hipStream_t s = stream ? (hipStream_t)stream : e->stream;
 	LIZEC_CHECK(hipSetDevice(e->device));
-	/* upload per-chunk tables into engine scratch (u64 slots reused) */
+	/* upload per-chunk tables into stream scratch (u64 slots reused):
+	 * [dptrs u64 x n][doffs u32 x n][coffs][counts] */
 	size_t words = (size_t)nchunks;               /* dptrs */
-	size_t meta = (words * 8 + 3 * words * 4 + 7) / 8 + words;
-	int r = ensure_ptrs(e, meta + 8);
+	size_t meta = words + (3 * words * 4 + 7) / 8;
+	size_t bytes = meta * 8;
+	lizec_stream_ctx *c;
+	int r = ctx_acquire(e, s, meta + 8, bytes, &c);
 	if (r != LIZEC_OK) return r;
-	uint64_t *d_ptrs = e->d_ptrs;
+	uint64_t *d_ptrs = c->d_ptrs;
 	uint32_t *d_doffs = (uint32_t *)(d_ptrs + nchunks);
 	uint32_t *d_coffs = d_doffs + nchunks;
 	uint32_t *d_counts = d_coffs + nchunks;
-	LIZEC_CHECK(hipMemcpyAsync(d_ptrs, chunk_dptrs, (size_t)nchunks * 8,
+	uint8_t *st = c->h_staging + 16 * 32 * 32;
+	memcpy(st, chunk_dptrs, (size_t)nchunks * 8);
+	memcpy(st + (size_t)nchunks * 8, data_offs, (size_t)nchunks * 4);
+	memcpy(st + (size_t)nchunks * 12, crc_offs, (size_t)nchunks * 4);
+	memcpy(st + (size_t)nchunks * 16, block_counts, (size_t)nchunks * 4);
+	LIZEC_CHECK(hipMemcpyAsync(d_ptrs, st, (size_t)nchunks * 20,
 	                           hipMemcpyHostToDevice, s));
-	LIZEC_CHECK(hipMemcpyAsync(d_doffs, data_offs, (size_t)nchunks * 4,
-	                           hipMemcpyHostToDevice, s));
-	LIZEC_CHECK(hipMemcpyAsync(d_coffs, crc_offs, (size_t)nchunks * 4,
-	                           hipMemcpyHostToDevice, s));
-	LIZEC_CHECK(hipMemcpyAsync(d_counts, block_counts, (size_t)nchunks * 4,
-	                           hipMemcpyHostToDevice, s));
+	LIZEC_CHECK(hipEventRecord(c->uploaded, s));
+	c->ev_recorded = true;
 	uint32_t max_blocks = 0;
 	for (int i = 0; i < nchunks; ++i)
 		if (block_counts[i] > max_blocks) max_blocks = block_counts[i];
