@@ -41,7 +41,7 @@ __global__ void fill_kernel(uint8_t *p, size_t n, uint32_t salt) {
 }
 
 template <int D, int CH, bool SWZ, bool NTST, bool NTLD = false,
-          bool TPIPE = false, bool QL = false>
+          bool TPIPE = false, bool QL = false, bool MR = false>
 static void launch_var(uint32_t part_len, int srcs, const uint8_t *tbls,
                        const uint64_t *src, const uint64_t *dst, int dests,
                        uint32_t tiles_per_part, uint32_t total_tiles,
@@ -49,7 +49,7 @@ static void launch_var(uint32_t part_len, int srcs, const uint8_t *tbls,
 	uint32_t grid = total_tiles < grid_cap ? total_tiles : grid_cap;
 	size_t lds = (size_t)D * srcs * (QL ? 16 : 32);
 	for (int base = 0; base + D <= dests; base += D)
-		hipLaunchKernelGGL(HIP_KERNEL_NAME(ec_encode_kernel<D, CH, SWZ, NTST, NTLD, TPIPE, QL>),
+		hipLaunchKernelGGL(HIP_KERNEL_NAME(ec_encode_kernel<D, CH, SWZ, NTST, NTLD, TPIPE, QL, MR>),
 		                   dim3(grid), dim3(kThreads), lds, s, part_len, srcs,
 		                   base, tbls, src, dst, dests, tiles_per_part,
 		                   total_tiles);
@@ -95,6 +95,23 @@ int main(int argc, char **argv) {
 	uint8_t *d_tbls_q;
 	CK(hipMalloc(&d_tbls_q, 16 * k * m));
 	CK(hipMemcpy(d_tbls_q, tbls_q, 16 * k * m, hipMemcpyHostToDevice));
+	/* packed mixed-radix (3+3+2) layout: 32 B padded per (dest,src);
+	 * derived from the ISA-L table by GF linearity over disjoint bits */
+	static uint8_t tbls_mr[32 * 32 * 32];
+	memset(tbls_mr, 0, sizeof(tbls_mr));
+	for (int i = 0; i < k * m; ++i) {
+		const uint8_t *t = tbls + i * 32;
+		uint8_t *q = tbls_mr + i * 32;
+		for (int v = 0; v < 8; ++v) {
+			q[v] = t[v];                                   /* c*v        */
+			q[8 + v] = t[(8 * v) & 15] ^ t[16 + (v >> 1)]; /* c*(v<<3)   */
+		}
+		for (int v = 0; v < 4; ++v)
+			q[16 + v] = t[16 + (v << 2)];                  /* c*(v<<6)   */
+	}
+	uint8_t *d_tbls_mr;
+	CK(hipMalloc(&d_tbls_mr, 32 * k * m));
+	CK(hipMemcpy(d_tbls_mr, tbls_mr, 32 * k * m, hipMemcpyHostToDevice));
 
 	std::vector<uint64_t> sp(stripes * k), dp(stripes * m);
 	for (int s = 0; s < stripes; ++s) {
@@ -130,9 +147,15 @@ int main(int argc, char **argv) {
 		int ch;
 		int d;
 		uint32_t grid_cap;
-		bool ql = false;
+		int tbl = 0;   /* 0 = full 32B, 1 = quarter-LUT 16B, 2 = MR 32B */
 	};
 	Cfg cfgs[] = {
+	    {"D2_CH4_mr    ", launch_var<2, 4, true, true, true, false, false, true>, 4, 2, 1048576, 2},
+	    {"D4_CH4_mr    ", launch_var<4, 4, true, true, true, false, false, true>, 4, 4, 1048576, 2},
+	    {"D4_CH2_mr    ", launch_var<4, 2, true, true, true, false, false, true>, 2, 4, 1048576, 2},
+	    {"D6_CH2_mr    ", launch_var<6, 2, true, true, true, false, false, true>, 2, 6, 1048576, 2},
+	    {"D6_CH4_mr    ", launch_var<6, 4, true, true, true, false, false, true>, 4, 6, 1048576, 2},
+	    {"D8_CH2_mr    ", launch_var<8, 2, true, true, true, false, false, true>, 2, 8, 1048576, 2},
 	    {"D1_CH4_swz_nt ", launch_var<1, 4, true, true>, 4, 1, 262144},
 	    {"D2_CH4_swz_nt ", launch_var<2, 4, true, true>, 4, 2, 262144},
 	    {"D2_CH4_swz_nt_ntld", launch_var<2, 4, true, true, true>, 4, 2, 262144},
@@ -178,7 +201,8 @@ int main(int argc, char **argv) {
 		uint32_t tot = tpp * stripes;
 		CK(hipMemset(d_par, 0, 4096));
 		/* warmup x2 */
-		const uint8_t *tb = c.ql ? d_tbls_q : d_tbls;
+		const uint8_t *tb = c.tbl == 1 ? d_tbls_q
+		                  : c.tbl == 2 ? d_tbls_mr : d_tbls;
 		for (int r = 0; r < 2; ++r)
 			c.fn((uint32_t)part_len, k, tb, d_sp, d_dp, m, tpp, tot,
 			     c.grid_cap, 0);

@@ -98,6 +98,47 @@ __device__ __forceinline__ void gf_macc_all_q(uint4 (&acc)[D][CH], int c,
 	}
 }
 
+/* Mixed-radix GF multiply: byte n = f0 + 8*f1 + 64*f2 (3+3+2-bit fields,
+ * disjoint bits => XOR over GF(2^8) linearity).  An 8-entry product table
+ * spans exactly the two source operands of one v_perm (sel 0-3 -> src1
+ * bytes, 4-7 -> src0), so a dest costs 3 perms + 3 xors per word instead
+ * of the quarter-LUT's 4+4 — at the price of a 32-byte padded LDS table
+ * per (src,dest) pair read as b128+b32.  Packed layout:
+ *   [0..3]   c*v, v=0..3        [4..7]   c*v, v=4..7
+ *   [8..11]  c*(v<<3), v=0..3   [12..15] c*(v<<3), v=4..7
+ *   [16..19] c*(v<<6), v=0..3   [20..31] pad */
+__device__ __forceinline__ uint32_t gf_macc_mr(uint32_t acc, const uint4 &T,
+                                               uint32_t t6, uint32_t f0,
+                                               uint32_t f1, uint32_t f2) {
+	uint32_t p0 = __builtin_amdgcn_perm(T.y, T.x, f0);
+	uint32_t p1 = __builtin_amdgcn_perm(T.w, T.z, f1);
+	uint32_t p2 = __builtin_amdgcn_perm(t6, t6, f2);
+	return acc ^ p0 ^ p1 ^ p2;
+}
+
+template <int D, int CH>
+__device__ __forceinline__ void gf_macc_all_mr(uint4 (&acc)[D][CH], int c,
+                                               const uint4 &w,
+                                               const uint4 (&T)[D],
+                                               const uint4 (&T6)[D]) {
+	constexpr uint32_t M3 = 0x07070707u, M2 = 0x03030303u;
+	uint32_t f[4][3];
+	const uint32_t ws[4] = {w.x, w.y, w.z, w.w};
+#pragma unroll
+	for (int q = 0; q < 4; ++q) {
+		f[q][0] = ws[q] & M3;
+		f[q][1] = (ws[q] >> 3) & M3;
+		f[q][2] = (ws[q] >> 6) & M2;
+	}
+#pragma unroll
+	for (int d = 0; d < D; ++d) {
+		acc[d][c].x = gf_macc_mr(acc[d][c].x, T[d], T6[d].x, f[0][0], f[0][1], f[0][2]);
+		acc[d][c].y = gf_macc_mr(acc[d][c].y, T[d], T6[d].x, f[1][0], f[1][1], f[1][2]);
+		acc[d][c].z = gf_macc_mr(acc[d][c].z, T[d], T6[d].x, f[2][0], f[2][1], f[2][2]);
+		acc[d][c].w = gf_macc_mr(acc[d][c].w, T[d], T6[d].x, f[3][0], f[3][1], f[3][2]);
+	}
+}
+
 /* Bijective XCD remap (8 XCDs): consecutive hardware block ids round-robin
  * the XCDs; remapped ids give each XCD one contiguous range. */
 __device__ __forceinline__ uint32_t xcd_remap(uint32_t b, uint32_t n) {
@@ -114,7 +155,7 @@ __device__ __forceinline__ uint4 ld_nt(const uint8_t *p) {
 }
 
 template <int D, int CH, bool SWZ, bool NTST, bool NTLD = false,
-          bool TPIPE = false, bool QL = false>
+          bool TPIPE = false, bool QL = false, bool MR = false>
 __global__ __launch_bounds__(kThreads) void ec_encode_kernel(
     uint32_t part_len, int srcs, int dest_base,
     const uint8_t *__restrict__ gftbls_dev,  /* 32*srcs*dests_total */
@@ -124,6 +165,7 @@ __global__ __launch_bounds__(kThreads) void ec_encode_kernel(
 	/* QL keeps its packed table in L only; H is sized 1 then, so the
 	 * TPIPE double-buffer path (H[1]) must not be combined with it. */
 	static_assert(!(QL && TPIPE), "QL and TPIPE are mutually exclusive");
+	static_assert(!(MR && (QL || TPIPE)), "MR excludes QL/TPIPE");
 	constexpr uint32_t kTile = kChunkBytes * CH;
 	extern __shared__ __attribute__((aligned(16))) uint8_t smem[];
 	const uint32_t tid = threadIdx.x;
@@ -168,6 +210,11 @@ __global__ __launch_bounds__(kThreads) void ec_encode_kernel(
 					if (QL) {
 						Lb[d] = *(const uint4 *)(smem +
 						                         ((size_t)d * srcs + j) * 16);
+					} else if (MR) {
+						const uint8_t *tb =
+						    smem + ((size_t)d * srcs + j) * 32;
+						Lb[d] = *(const uint4 *)tb;
+						Hb[d].x = *(const uint32_t *)(tb + 16);
 					} else {
 						const uint8_t *tb =
 						    smem + ((size_t)d * srcs + j) * 32;
@@ -202,6 +249,8 @@ __global__ __launch_bounds__(kThreads) void ec_encode_kernel(
 				for (int c = 0; c < CH; ++c) {
 					if (QL)
 						gf_macc_all_q<D, CH>(acc, c, w[c], Lc);
+					else if (MR)
+						gf_macc_all_mr<D, CH>(acc, c, w[c], Lc, Hc);
 					else
 						gf_macc_all<D, CH>(acc, c, w[c], Lc, Hc);
 				}
@@ -242,6 +291,11 @@ __global__ __launch_bounds__(kThreads) void ec_encode_kernel(
 					if (QL) {
 						L[d] = *(const uint4 *)(smem +
 						                        ((size_t)d * srcs + j) * 16);
+					} else if (MR) {
+						const uint8_t *tb =
+						    smem + ((size_t)d * srcs + j) * 32;
+						L[d] = *(const uint4 *)tb;
+						H[d].x = *(const uint32_t *)(tb + 16);
 					} else {
 						const uint8_t *tb =
 						    smem + ((size_t)d * srcs + j) * 32;
@@ -256,6 +310,8 @@ __global__ __launch_bounds__(kThreads) void ec_encode_kernel(
 						uint4 w = *(const uint4 *)(sp + off);
 						if (QL)
 							gf_macc_all_q<D, CH>(acc, c, w, L);
+						else if (MR)
+							gf_macc_all_mr<D, CH>(acc, c, w, L, H);
 						else
 							gf_macc_all<D, CH>(acc, c, w, L, H);
 					}
