@@ -156,12 +156,18 @@ def solve_fold_constant(targets):
     return K
 
 
+_fc_cache = {}
+
+
 def fold_constants(nacc):
     """(KL, KH) for fold distance nacc*16 bytes."""
+    if nacc in _fc_cache:
+        return _fc_cache[nacc]
     adv = lambda s: rawcrc(b"\x00" * (16 * nacc), s)
     cn = [rawcrc16(1 << n) for n in range(128)]
     KL = solve_fold_constant([adv(cn[i]) for i in range(64)])
     KH = solve_fold_constant([adv(cn[64 + i]) for i in range(64)])
+    _fc_cache[nacc] = (KL, KH)
     return KL, KH
 
 
