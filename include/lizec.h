@@ -223,6 +223,33 @@ int lizec_scrub_batch_strided(lizec_engine *e, const uint64_t *chunk_dptrs,
                               uint32_t block_stride, uint32_t crc_stride,
                               int32_t *dev_status_out, void *stream);
 
+/* Pinned host memory for the streaming APIs (hipHostMalloc/hipHostFree):
+ * pageable buffers work too but degrade the pipeline to synchronous
+ * copies. */
+int lizec_host_alloc(void **ptr, uint64_t bytes);
+void lizec_host_free(void *p);
+
+/* Streaming chunk rebuild — ChunkReplicator::replicate
+ * (chunk_replicator.cc:139-196) as a host-to-host pipeline: H2D the
+ * surviving parts, EC-recover the erased parts, CRC every recovered
+ * 64 KiB block (chunk_replicator.cc:189), assemble complete MooseFS part
+ * images (chunk.cc:126-188: signature bytes at 0, big-endian CRC array at
+ * crc_off, blocks at header_size) and D2H them — double-buffered over two
+ * streams so H2D/compute/D2H overlap.
+ *  part_len    : bytes per part, multiple of 65536
+ *  ic, oc      : surviving parts in / parts to rebuild per chunk
+ *  gftbls      : 32*ic*oc recover tables (lizec_rs_tables)
+ *  host_src    : nchunks*ic HOST addresses of surviving part bytes
+ *  sigs        : nchunks*oc signatures (sig_len bytes each), or NULL
+ *  header_size : image header bytes; crc_off: CRC array offset within it
+ *  host_dst    : nchunks*oc HOST addresses, header_size+part_len each
+ *  sub_batch   : chunks per pipeline stage (0 = auto, ~1 GiB staging) */
+int lizec_replicate_run(lizec_engine *e, uint64_t part_len, int ic, int oc,
+                        const uint8_t *gftbls, const uint64_t *host_src,
+                        const uint8_t *sigs, uint32_t sig_len,
+                        uint32_t header_size, uint32_t crc_off,
+                        const uint64_t *host_dst, int nchunks, int sub_batch);
+
 #ifdef __cplusplus
 }
 #endif

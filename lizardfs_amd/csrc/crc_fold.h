@@ -124,12 +124,20 @@ __device__ __forceinline__ uint32_t crc_reduce16(const uint32_t (&acc)[4],
 }
 
 /* 16-byte load; AL16=false uses dword loads for 4-mod-16 bases (the
- * INTERLEAVED chunk format puts block data at offset 4). */
-template <bool AL16>
+ * INTERLEAVED chunk format puts block data at offset 4); NT streams
+ * through non-temporal loads (read-once data). */
+template <bool AL16, bool NT = false>
 __device__ __forceinline__ uint4 crc_ld16(const uint8_t *p) {
-	if constexpr (AL16) return *(const uint4 *)p;
-	const uint32_t *u = (const uint32_t *)p;
-	return make_uint4(u[0], u[1], u[2], u[3]);
+	if constexpr (AL16 && NT) {
+		typedef unsigned int u32x4 __attribute__((ext_vector_type(4)));
+		u32x4 v = __builtin_nontemporal_load((const u32x4 *)p);
+		return make_uint4(v.x, v.y, v.z, v.w);
+	} else if constexpr (AL16) {
+		return *(const uint4 *)p;
+	} else {
+		const uint32_t *u = (const uint32_t *)p;
+		return make_uint4(u[0], u[1], u[2], u[3]);
+	}
 }
 
 /* Per-wave CRC of one block via carry-less folding.
@@ -142,7 +150,7 @@ __device__ __forceinline__ uint4 crc_ld16(const uint8_t *p) {
  * a shfl tree with the advance matrices; lane 0 splices the C span CRCs.
  * Host guarantees block_len % (C * 64 * 16 * BV) == 0 (BV = 8, or 4 when
  * C = 4).  Returns the block CRC (lane 0's value is authoritative). */
-template <int C, int NACC, bool AL16 = true>
+template <int C, int NACC, bool AL16 = true, bool NT = false>
 __device__ uint32_t crc_block_wave_fold(const uint8_t *__restrict__ block,
                                         uint32_t block_len, uint32_t seed,
                                         const uint32_t *T0,
@@ -159,7 +167,7 @@ __device__ uint32_t crc_block_wave_fold(const uint8_t *__restrict__ block,
 	for (int c = 0; c < C; ++c)
 #pragma unroll
 		for (int q = 0; q < BV; ++q)
-			w[c][q] = crc_ld16<AL16>(base + c * span + q * 16);
+			w[c][q] = crc_ld16<AL16, NT>(base + c * span + q * 16);
 #pragma unroll
 	for (int c = 0; c < C; ++c) {
 		const uint32_t raw0 =
@@ -180,7 +188,7 @@ __device__ uint32_t crc_block_wave_fold(const uint8_t *__restrict__ block,
 		for (int c = 0; c < C; ++c)
 #pragma unroll
 			for (int q = 0; q < BV; ++q)
-				w[c][q] = crc_ld16<AL16>(base + c * span + i + q * 16);
+				w[c][q] = crc_ld16<AL16, NT>(base + c * span + i + q * 16);
 #pragma unroll
 		for (int q = 0; q < BV; ++q)
 #pragma unroll

@@ -256,7 +256,7 @@ __global__ __launch_bounds__(kThreads) void crc32_blocks_kernel_multi(
  * LDS holds only the epilogue byte table T0 and the combine matrices. */
 constexpr int kCrcFoldLdsWords = 256 + kCrcMatCount * 32;
 
-template <int C, int NACC, bool AL16 = true>
+template <int C, int NACC, bool AL16 = true, bool NT = false>
 __global__ __launch_bounds__(kThreads) void crc32_blocks_kernel_fold(
     const uint8_t *__restrict__ buf, uint32_t block_len, uint64_t nblocks,
     uint32_t seed, const uint32_t *__restrict__ crc_const,
@@ -271,18 +271,25 @@ __global__ __launch_bounds__(kThreads) void crc32_blocks_kernel_fold(
 	const int lane = threadIdx.x & 63;
 	for (uint64_t blk = (uint64_t)blockIdx.x * 4 + wave; blk < nblocks;
 	     blk += (uint64_t)gridDim.x * 4) {
-		uint32_t crc = crc_block_wave_fold<C, NACC, AL16>(
+		uint32_t crc = crc_block_wave_fold<C, NACC, AL16, NT>(
 		    buf + blk * block_len, block_len, seed, stabs, stabs + 256, lane);
 		if (lane == 0) out[blk] = crc;
 	}
 }
 
-/* Batched chunk scrub — hdd_int_test semantics (hddspacemgr.cc:2148-2212):
- * for every 64 KiB block of every MooseFS-format chunk-part image, compare
- * mycrc32(0, block, MFSBLOCKSIZE) against the stored CRC array entry
- * (big-endian u32 at crc_off + 4*b, cf. get32bit at hddspacemgr.cc:2183 and
- * chunk.cc:183-188).  status[c] collects the FIRST damaged block index via
- * atomicMin (host pre-fills INT32_MAX = clean). */
+/* Batched chunk scrub / image-CRC fill.
+ *
+ * WRITE=false: hdd_int_test semantics (hddspacemgr.cc:2148-2212): for
+ * every 64 KiB block of every chunk-part image, compare mycrc32(0, block,
+ * MFSBLOCKSIZE) against the stored CRC array entry (big-endian u32 at
+ * crc_off + crc_stride*b, cf. get32bit at hddspacemgr.cc:2183 and
+ * chunk.cc:183-188).  status[c] collects the FIRST damaged block index
+ * via atomicMin (host pre-fills INT32_MAX = clean).
+ *
+ * WRITE=true: image assembly for the replication pipeline
+ * (chunk_replicator.cc:189 + chunk.cc:126-188): compute the same CRCs and
+ * STORE them big-endian into the image's CRC array (status unused). */
+template <bool WRITE>
 __global__ __launch_bounds__(kThreads) void scrub_chunks_kernel(
     const uint64_t *__restrict__ chunk_dptrs,
     const uint32_t *__restrict__ data_offs,
@@ -312,15 +319,23 @@ __global__ __launch_bounds__(kThreads) void scrub_chunks_kernel(
 		 * instantiation there (uint4 loads would be misaligned UB) */
 		uint32_t crc =
 		    (((uintptr_t)blockp & 15) == 0)
-		        ? crc_block_wave_fold<2, 2, true>(blockp, 65536u, 0u, T0,
+		        ? crc_block_wave_fold<1, 2, true>(blockp, 65536u, 0u, T0,
 		                                          mats, lane)
-		        : crc_block_wave_fold<2, 2, false>(blockp, 65536u, 0u, T0,
+		        : crc_block_wave_fold<1, 2, false>(blockp, 65536u, 0u, T0,
 		                                           mats, lane);
 		if (lane == 0) {
-			const uint8_t *p = img + crc_offs[c] + crc_stride * b;
-			uint32_t stored = ((uint32_t)p[0] << 24) | ((uint32_t)p[1] << 16) |
-			                  ((uint32_t)p[2] << 8) | p[3];
-			if (stored != crc) atomicMin(&status[c], (int32_t)b);
+			uint8_t *p = (uint8_t *)img + crc_offs[c] + crc_stride * b;
+			if (WRITE) {
+				p[0] = (uint8_t)(crc >> 24);
+				p[1] = (uint8_t)(crc >> 16);
+				p[2] = (uint8_t)(crc >> 8);
+				p[3] = (uint8_t)crc;
+			} else {
+				uint32_t stored = ((uint32_t)p[0] << 24) |
+				                  ((uint32_t)p[1] << 16) |
+				                  ((uint32_t)p[2] << 8) | p[3];
+				if (stored != crc) atomicMin(&status[c], (int32_t)b);
+			}
 		}
 	}
 }
@@ -687,53 +702,39 @@ extern "C" int lizec_crc32_batch(lizec_engine *e, const void *dev_buf,
 	const char *sl = getenv("LIZEC_CRC_SLICE");
 	const char *im = getenv("LIZEC_CRC_IMPL");
 	const char *fn = getenv("LIZEC_CRC_FOLD_NACC");
-	int chains = ch ? atoi(ch) : 2;   /* C=4 measured worse: half-line bursts re-thrash L1 (profiles) */
+	const char *nt = getenv("LIZEC_CRC_NT");
+	/* fold C=1 NACC=2 measured best (r2a: 4779 GB/s vs table 3474) */
+	int chains = ch ? atoi(ch) : 1;
 	int slice = sl ? atoi(sl) : 8;   /* slice-16 measured -16% within-box (profiles) */
 	bool fold = !(im && strcmp(im, "table") == 0);
 	int nacc = fn ? atoi(fn) : 2;
+	bool ntld = nt && atoi(nt) != 0;
 	bool al16 = (((uintptr_t)dev_buf | block_len) & 15) == 0;
 	/* carry-less-folding path (default): block must split into C spans of
-	 * whole 64-lane x BV*16-byte bursts */
-	if (fold && block_len % (2 * 64 * 16 * 8) == 0 && chains == 2) {
+	 * whole 64-lane x BV*16-byte bursts (BV=8) */
+	if (fold && block_len % ((uint32_t)chains * 64 * 16 * 8) == 0 &&
+	    (chains == 1 || chains == 2)) {
 		const uint8_t *b = (const uint8_t *)dev_buf;
-#define LIZEC_LAUNCH_FOLD(C, N, A)                                          \
-	hipLaunchKernelGGL(HIP_KERNEL_NAME(crc32_blocks_kernel_fold<C, N, A>),  \
-	                   dim3(grid), dim3(kThreads), 0, s, b, block_len,      \
+#define LIZEC_LAUNCH_FOLD(C, N, A, NT)                                       \
+	hipLaunchKernelGGL(HIP_KERNEL_NAME(crc32_blocks_kernel_fold<C, N, A, NT>), \
+	                   dim3(grid), dim3(kThreads), 0, s, b, block_len,       \
 	                   nblocks, seed, e->d_crc_const, dev_crcs_out)
-		if (al16) {
-			if (nacc == 1) LIZEC_LAUNCH_FOLD(2, 1, true);
-			else if (nacc == 4) LIZEC_LAUNCH_FOLD(2, 4, true);
-			else LIZEC_LAUNCH_FOLD(2, 2, true);
+		if (!al16) {
+			if (chains == 2) LIZEC_LAUNCH_FOLD(2, 2, false, false);
+			else LIZEC_LAUNCH_FOLD(1, 2, false, false);
+		} else if (chains == 2) {
+			if (nacc == 1) LIZEC_LAUNCH_FOLD(2, 1, true, false);
+			else LIZEC_LAUNCH_FOLD(2, 2, true, false);
+		} else if (ntld) {
+			if (nacc == 1) LIZEC_LAUNCH_FOLD(1, 1, true, true);
+			else if (nacc == 4) LIZEC_LAUNCH_FOLD(1, 4, true, true);
+			else LIZEC_LAUNCH_FOLD(1, 2, true, true);
 		} else {
-			LIZEC_LAUNCH_FOLD(2, 2, false);
+			if (nacc == 1) LIZEC_LAUNCH_FOLD(1, 1, true, false);
+			else if (nacc == 4) LIZEC_LAUNCH_FOLD(1, 4, true, false);
+			else LIZEC_LAUNCH_FOLD(1, 2, true, false);
 		}
 #undef LIZEC_LAUNCH_FOLD
-		LIZEC_CHECK(hipGetLastError());
-		return LIZEC_OK;
-	}
-	if (fold && block_len % (4 * 64 * 16 * 4) == 0 && chains == 4) {
-		const uint8_t *b = (const uint8_t *)dev_buf;
-		if (al16)
-			hipLaunchKernelGGL(HIP_KERNEL_NAME(crc32_blocks_kernel_fold<4, 1, true>),
-			                   dim3(grid), dim3(kThreads), 0, s, b, block_len,
-			                   nblocks, seed, e->d_crc_const, dev_crcs_out);
-		else
-			hipLaunchKernelGGL(HIP_KERNEL_NAME(crc32_blocks_kernel_fold<4, 1, false>),
-			                   dim3(grid), dim3(kThreads), 0, s, b, block_len,
-			                   nblocks, seed, e->d_crc_const, dev_crcs_out);
-		LIZEC_CHECK(hipGetLastError());
-		return LIZEC_OK;
-	}
-	if (fold && block_len % (64 * 16 * 8) == 0 && chains == 1) {
-		const uint8_t *b = (const uint8_t *)dev_buf;
-		if (al16)
-			hipLaunchKernelGGL(HIP_KERNEL_NAME(crc32_blocks_kernel_fold<1, 2, true>),
-			                   dim3(grid), dim3(kThreads), 0, s, b, block_len,
-			                   nblocks, seed, e->d_crc_const, dev_crcs_out);
-		else
-			hipLaunchKernelGGL(HIP_KERNEL_NAME(crc32_blocks_kernel_fold<1, 2, false>),
-			                   dim3(grid), dim3(kThreads), 0, s, b, block_len,
-			                   nblocks, seed, e->d_crc_const, dev_crcs_out);
 		LIZEC_CHECK(hipGetLastError());
 		return LIZEC_OK;
 	}
@@ -800,10 +801,28 @@ extern "C" int lizec_scrub_batch_strided(
 	uint64_t total = (uint64_t)nchunks * max_blocks;
 	uint64_t groups = (total + 3) / 4;
 	uint32_t grid = (uint32_t)(groups < 131072 ? groups : 131072);
-	hipLaunchKernelGGL(scrub_chunks_kernel, dim3(grid), dim3(kThreads), 0, s,
+	hipLaunchKernelGGL(HIP_KERNEL_NAME(scrub_chunks_kernel<false>),
+	                   dim3(grid), dim3(kThreads), 0, s,
 	                   d_ptrs, d_doffs, d_coffs, d_counts, (uint32_t)nchunks,
 	                   max_blocks, block_stride, crc_stride, e->d_crc_const,
 	                   dev_status_out);
+	LIZEC_CHECK(hipGetLastError());
+	return LIZEC_OK;
+}
+
+/* Launch the WRITE variant on already-device-resident meta arrays (used by
+ * the replication pipeline, which stages meta itself). */
+static int image_crc_launch(lizec_engine *e, const uint64_t *d_ptrs,
+                            const uint32_t *d_doffs, const uint32_t *d_coffs,
+                            const uint32_t *d_counts, int nimages,
+                            uint32_t max_blocks, hipStream_t s) {
+	uint64_t total = (uint64_t)nimages * max_blocks;
+	uint64_t groups = (total + 3) / 4;
+	uint32_t grid = (uint32_t)(groups < 131072 ? groups : 131072);
+	hipLaunchKernelGGL(HIP_KERNEL_NAME(scrub_chunks_kernel<true>),
+	                   dim3(grid), dim3(kThreads), 0, s, d_ptrs, d_doffs,
+	                   d_coffs, d_counts, (uint32_t)nimages, max_blocks,
+	                   65536u, 4u, e->d_crc_const, nullptr);
 	LIZEC_CHECK(hipGetLastError());
 	return LIZEC_OK;
 }
@@ -817,4 +836,207 @@ extern "C" int lizec_scrub_batch(lizec_engine *e, const uint64_t *chunk_dptrs,
 	return lizec_scrub_batch_strided(e, chunk_dptrs, data_offs, crc_offs,
 	                                 block_counts, nchunks, 65536u, 4u,
 	                                 dev_status_out, stream);
+}
+
+/* ------------------------------------------------------------------ */
+/* Streaming replication pipeline                  */
+/* ------------------------------------------------------------------ */
+
+extern "C" int lizec_host_alloc(void **ptr, uint64_t bytes) {
+	*ptr = nullptr;
+	LIZEC_CHECK(hipHostMalloc(ptr, bytes));
+	return LIZEC_OK;
+}
+
+extern "C" void lizec_host_free(void *p) {
+	(void)hipHostFree(p);
+}
+
+/* One double-buffer slot of the replication pipeline. */
+struct repl_slot {
+	hipStream_t stream = nullptr;
+	hipEvent_t done = nullptr;
+	uint8_t *d_in = nullptr;    /* B * ic * part_len           */
+	uint8_t *d_img = nullptr;   /* B * oc * img_bytes          */
+	uint64_t *d_meta = nullptr; /* ptr tables + image meta     */
+	uint8_t *h_stage = nullptr; /* pinned: meta + signatures   */
+	bool used = false;
+};
+
+static void repl_slot_free(repl_slot &sl) {
+	(void)hipFree(sl.d_in);
+	(void)hipFree(sl.d_img);
+	(void)hipFree(sl.d_meta);
+	(void)hipHostFree(sl.h_stage);
+	if (sl.done) (void)hipEventDestroy(sl.done);
+	if (sl.stream) (void)hipStreamDestroy(sl.stream);
+	sl = repl_slot();
+}
+
+/* The ChunkReplicator::replicate loop (chunk_replicator.cc:139-196) as a
+ * host-to-host streaming pipeline: pull surviving parts from host memory
+ * (the reference pulls them from peer sockets), recover the erased parts
+ * on-GPU, CRC every recovered 64 KiB block (chunk_replicator.cc:189), and
+ * emit complete MooseFS part images (chunk.cc:126-188 geometry: caller's
+ * signature bytes at 0, BE CRC array at crc_off, blocks at header_size)
+ * into caller host buffers — H2D, compute and D2H double-buffered on two
+ * streams.  Use lizec_host_alloc'd (pinned) buffers for real overlap;
+ * pageable buffers degrade to synchronous copies but stay correct.
+ *
+ * host_src:  nchunks*ic host addresses (surviving part bytes, part_len each)
+ * gftbls:    32*ic*oc recover tables (lizec_rs_tables)
+ * sigs:      nchunks*oc signatures, sig_len bytes each
+ * host_dst:  nchunks*oc host addresses (header_size + part_len each)
+ * sub_batch: chunks per pipeline stage (0 = auto) */
+extern "C" int lizec_replicate_run(lizec_engine *e, uint64_t part_len,
+                                   int ic, int oc, const uint8_t *gftbls,
+                                   const uint64_t *host_src,
+                                   const uint8_t *sigs, uint32_t sig_len,
+                                   uint32_t header_size, uint32_t crc_off,
+                                   const uint64_t *host_dst, int nchunks,
+                                   int sub_batch) {
+	if (!e || !gftbls || !host_src || !host_dst || nchunks < 1)
+		return LIZEC_EINVAL;
+	if (ic < 1 || ic > 32 || oc < 1 || oc > 32) return LIZEC_EINVAL;
+	if (part_len == 0 || part_len % 65536 || part_len > (uint64_t)1 << 31)
+		return LIZEC_EINVAL;
+	if (sig_len > header_size || crc_off + 4 * (part_len / 65536) > header_size)
+		return LIZEC_EINVAL;
+	LIZEC_CHECK(hipSetDevice(e->device));
+
+	const uint64_t img_bytes = header_size + part_len;
+	int B = sub_batch;
+	if (B <= 0) {
+		/* ~1 GiB of device staging per slot */
+		uint64_t per_chunk = (uint64_t)ic * part_len + oc * img_bytes;
+		B = (int)(((uint64_t)1 << 30) / per_chunk);
+		if (B < 1) B = 1;
+		if (B > 64) B = 64;
+	}
+	if (B > nchunks) B = nchunks;
+
+	/* per-(stripe,part) device pointer tables + per-image CRC meta:
+	 * layout in d_meta/h_stage (8-byte aligned blocks):
+	 *   [0)              src ptrs   B*ic u64
+	 *   [src_end)        dst ptrs   B*oc u64
+	 *   [meta_img)       image ptrs B*oc u64
+	 *   [meta_off)       doffs/coffs/counts  3 * B*oc u32  */
+	const size_t n_src = (size_t)B * ic, n_dst = (size_t)B * oc;
+	const size_t meta_words = n_src + n_dst + n_dst + (3 * n_dst * 4 + 7) / 8;
+	const size_t stage_bytes = meta_words * 8 + (size_t)B * oc * sig_len;
+
+	uint8_t *d_tbl = nullptr;
+	repl_slot sl[2];
+	int rc = LIZEC_OK;
+	uint8_t packed[16 * 32 * 32];
+	for (int i = 0; i < ic * oc; ++i)
+		pack_quarter_lut(gftbls + (size_t)i * 32, packed + (size_t)i * 16);
+	if (hipMalloc(&d_tbl, (size_t)16 * ic * oc) != hipSuccess)
+		return LIZEC_ENOMEM;
+	if (hipMemcpy(d_tbl, packed, (size_t)16 * ic * oc,
+	              hipMemcpyHostToDevice) != hipSuccess) {
+		(void)hipFree(d_tbl);
+		return LIZEC_EHIP;
+	}
+	for (int i = 0; i < 2 && rc == LIZEC_OK; ++i) {
+		if (hipStreamCreate(&sl[i].stream) != hipSuccess ||
+		    hipEventCreateWithFlags(&sl[i].done, hipEventDisableTiming) !=
+		        hipSuccess ||
+		    hipMalloc(&sl[i].d_in, (size_t)B * ic * part_len) != hipSuccess ||
+		    hipMalloc(&sl[i].d_img, (size_t)B * oc * img_bytes) !=
+		        hipSuccess ||
+		    hipMalloc(&sl[i].d_meta, meta_words * 8) != hipSuccess ||
+		    hipHostMalloc(&sl[i].h_stage, stage_bytes) != hipSuccess)
+			rc = LIZEC_ENOMEM;
+	}
+
+	for (int c0 = 0; c0 < nchunks && rc == LIZEC_OK; c0 += B) {
+		int nb = nchunks - c0 < B ? nchunks - c0 : B;
+		repl_slot &s = sl[(c0 / B) & 1];
+		hipStream_t st = s.stream;
+		if (s.used) {
+			if (hipEventSynchronize(s.done) != hipSuccess) {
+				rc = LIZEC_EHIP;
+				break;
+			}
+		}
+		s.used = true;
+		/* stage pointer tables + sigs (host side; slot is idle now) */
+		uint64_t *h_src = (uint64_t *)s.h_stage;
+		uint64_t *h_dst = h_src + n_src;
+		uint64_t *h_img = h_dst + n_dst;
+		uint32_t *h_off = (uint32_t *)(h_img + n_dst);
+		uint32_t *h_coff = h_off + n_dst;
+		uint32_t *h_cnt = h_coff + n_dst;
+		uint8_t *h_sig = s.h_stage + meta_words * 8;
+		const uint32_t nblocks = (uint32_t)(part_len / 65536);
+		for (int i = 0; i < nb; ++i)
+			for (int j = 0; j < ic; ++j)
+				h_src[i * ic + j] = (uint64_t)(s.d_in +
+				                               ((size_t)i * ic + j) * part_len);
+		for (int i = 0; i < nb; ++i)
+			for (int j = 0; j < oc; ++j) {
+				size_t n = (size_t)i * oc + j;
+				uint64_t img = (uint64_t)(s.d_img + n * img_bytes);
+				h_img[n] = img;
+				h_dst[n] = img + header_size;
+				h_off[n] = header_size;
+				h_coff[n] = crc_off;
+				h_cnt[n] = nblocks;
+			}
+		if (sigs)
+			memcpy(h_sig, sigs + (size_t)c0 * oc * sig_len,
+			       (size_t)nb * oc * sig_len);
+		/* enqueue the stage: H2D parts, header init, recover, CRC, D2H */
+		for (int i = 0; i < nb && rc == LIZEC_OK; ++i)
+			for (int j = 0; j < ic; ++j)
+				if (hipMemcpyAsync(s.d_in + ((size_t)i * ic + j) * part_len,
+				                   (const void *)host_src[(size_t)(c0 + i) *
+				                                          ic + j],
+				                   part_len, hipMemcpyHostToDevice,
+				                   st) != hipSuccess)
+					rc = LIZEC_EHIP;
+		if (rc != LIZEC_OK) break;
+		if (hipMemcpyAsync(s.d_meta, s.h_stage, meta_words * 8,
+		                   hipMemcpyHostToDevice, st) != hipSuccess) {
+			rc = LIZEC_EHIP;
+			break;
+		}
+		for (int n = 0; n < nb * oc && rc == LIZEC_OK; ++n) {
+			uint8_t *img = s.d_img + (size_t)n * img_bytes;
+			if (hipMemsetAsync(img, 0, header_size, st) != hipSuccess)
+				rc = LIZEC_EHIP;
+			else if (sigs && sig_len &&
+			         hipMemcpyAsync(img, h_sig + (size_t)n * sig_len, sig_len,
+			                        hipMemcpyHostToDevice, st) != hipSuccess)
+				rc = LIZEC_EHIP;
+		}
+		if (rc != LIZEC_OK) break;
+		uint64_t *d_src = s.d_meta;
+		uint64_t *d_dst = s.d_meta + n_src;
+		uint64_t *d_img_ptrs = d_dst + n_dst;
+		uint32_t *d_off = (uint32_t *)(d_img_ptrs + n_dst);
+		uint32_t *d_coff = d_off + n_dst;
+		uint32_t *d_cnt = d_coff + n_dst;
+		rc = run_batch(part_len, ic, oc, d_tbl, d_src, d_dst, (uint32_t)nb,
+		               st);
+		if (rc != LIZEC_OK) break;
+		rc = image_crc_launch(e, d_img_ptrs, d_off, d_coff, d_cnt, nb * oc,
+		                      nblocks, st);
+		if (rc != LIZEC_OK) break;
+		for (int n = 0; n < nb * oc && rc == LIZEC_OK; ++n)
+			if (hipMemcpyAsync((void *)host_dst[(size_t)c0 * oc + n],
+			                   s.d_img + (size_t)n * img_bytes, img_bytes,
+			                   hipMemcpyDeviceToHost, st) != hipSuccess)
+				rc = LIZEC_EHIP;
+		if (rc != LIZEC_OK) break;
+		if (hipEventRecord(s.done, st) != hipSuccess) rc = LIZEC_EHIP;
+	}
+	for (int i = 0; i < 2; ++i) {
+		if (sl[i].used && sl[i].stream)
+			(void)hipStreamSynchronize(sl[i].stream);
+		repl_slot_free(sl[i]);
+	}
+	(void)hipFree(d_tbl);
+	return rc;
 }

@@ -97,6 +97,16 @@ def lib():
         ctypes.POINTER(ctypes.c_uint32), ctypes.POINTER(ctypes.c_uint32),
         ctypes.POINTER(ctypes.c_uint32), ctypes.c_int, ctypes.c_uint32,
         ctypes.c_uint32, ctypes.c_void_p, ctypes.c_void_p]
+    L.lizec_host_alloc.restype = ctypes.c_int
+    L.lizec_host_alloc.argtypes = [ctypes.POINTER(ctypes.c_void_p),
+                                   ctypes.c_uint64]
+    L.lizec_host_free.argtypes = [ctypes.c_void_p]
+    L.lizec_replicate_run.restype = ctypes.c_int
+    L.lizec_replicate_run.argtypes = [
+        ctypes.c_void_p, ctypes.c_uint64, ctypes.c_int, ctypes.c_int, u8p,
+        ctypes.POINTER(ctypes.c_uint64), u8p, ctypes.c_uint32,
+        ctypes.c_uint32, ctypes.c_uint32, ctypes.POINTER(ctypes.c_uint64),
+        ctypes.c_int, ctypes.c_int]
     _lib = L
     return _lib
 
@@ -110,6 +120,34 @@ def check(code, what=""):
 
 def gpu_count():
     return lib().lizec_gpu_count()
+
+
+def pinned_empty(shape, dtype=None):
+    """numpy array backed by pinned (hipHostMalloc) memory — gives the
+    streaming pipeline (lizec_replicate_run) true async copies.  The
+    allocation is freed when the returned array is garbage-collected."""
+    import numpy as np
+    dtype = np.dtype(dtype or np.uint8)
+    n = int(np.prod(shape)) * dtype.itemsize
+    p = ctypes.c_void_p()
+    check(lib().lizec_host_alloc(ctypes.byref(p), max(n, 1)),
+          "lizec_host_alloc")
+    buf = (ctypes.c_uint8 * n).from_address(p.value)
+    arr = np.frombuffer(buf, dtype=dtype).reshape(shape)
+    # tie the allocation's lifetime to the array
+    arr.base.base._lizec_finalizer = _HostMem(p)  # type: ignore[attr-defined]
+    return arr
+
+
+class _HostMem:
+    def __init__(self, p):
+        self._p = p
+
+    def __del__(self):
+        try:
+            lib().lizec_host_free(self._p)
+        except Exception:
+            pass
 
 
 _engines = {}

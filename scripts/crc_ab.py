@@ -22,13 +22,14 @@ BLOCK = 65536
 REPS = 10
 
 CFGS = [
-    # name, impl, chains, nacc
-    ("table_c2_s8", "table", "2", ""),
-    ("fold_c2_n1 ", "fold", "2", "1"),
-    ("fold_c2_n2 ", "fold", "2", "2"),
-    ("fold_c2_n4 ", "fold", "2", "4"),
-    ("fold_c1_n2 ", "fold", "1", "2"),
-    ("fold_c4_n1 ", "fold", "4", "1"),
+    # name, impl, chains, nacc, nt
+    ("table_c2_s8", "table", "2", "", "0"),
+    ("fold_c1_n1 ", "fold", "1", "1", "0"),
+    ("fold_c1_n2 ", "fold", "1", "2", "0"),
+    ("fold_c1_n4 ", "fold", "1", "4", "0"),
+    ("fold_c1_n1nt", "fold", "1", "1", "1"),
+    ("fold_c1_n2nt", "fold", "1", "2", "1"),
+    ("fold_c2_n2 ", "fold", "2", "2", "0"),
 ]
 
 
@@ -43,10 +44,11 @@ def main():
     ref = None
     e0 = torch.cuda.Event(enable_timing=True)
     e1 = torch.cuda.Event(enable_timing=True)
-    for name, impl, chains, nacc in CFGS:
+    for name, impl, chains, nacc, nt in CFGS:
         os.environ["LIZEC_CRC_IMPL"] = impl
         os.environ["LIZEC_CRC_CHAINS"] = chains
         os.environ["LIZEC_CRC_FOLD_NACC"] = nacc or "2"
+        os.environ["LIZEC_CRC_NT"] = nt
         out.zero_()
         torch.cuda.synchronize()
         for _ in range(3):
@@ -70,7 +72,8 @@ def main():
 
     # seeded + odd-size sanity through the generic path stays correct
     os.environ["LIZEC_CRC_IMPL"] = "fold"
-    os.environ["LIZEC_CRC_CHAINS"] = "2"
+    os.environ["LIZEC_CRC_CHAINS"] = "1"
+    os.environ["LIZEC_CRC_NT"] = "0"
     small = buf[:3 * 1024 * 1024]
     c1 = lcrc.crc32_blocks(small, 3072, seed=0xDEADBEEF)
     os.environ["LIZEC_CRC_IMPL"] = "table"
