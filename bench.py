@@ -248,6 +248,10 @@ def main():
                     help="gloo allows exercising the multi-rank path on a "
                          "single GPU (coordination on CPU; compute on GPU)")
     ap.add_argument("--skip-cpu-baseline", action="store_true")
+    ap.add_argument("--validate", action="store_true",
+                    help="after timing, bit-check recovered parts against "
+                         "the original data (decode/mixed round-trip) and "
+                         "the all-gather's own-rank slot")
     args = ap.parse_args()
 
     K, M, ERASED, S_default = OPS[args.op]
@@ -361,8 +365,9 @@ def main():
         outs = {i: torch.empty((S, PART_LEN), dtype=torch.uint8,
                                device="cuda") for i in ERASED}
         if distributed:
+            gdev = "cuda" if args.backend == "nccl" else "cpu"
             gathered = [
-                [torch.empty((S, PART_LEN), dtype=torch.uint8, device="cuda")
+                [torch.empty((S, PART_LEN), dtype=torch.uint8, device=gdev)
                  for _ in range(world)] for _ in ERASED]
 
         def step():
@@ -370,10 +375,13 @@ def main():
             rs.recover_batch(frags, erased=erased_full, want=set(ERASED),
                              out=outs)
             if distributed:
-                # RCCL all-gather of recovered parts over xGMI so every
-                # rank holds the full recovered set
+                # all-gather of recovered parts so every rank holds the
+                # full recovered set — RCCL over xGMI under
+                # nccl; under gloo (single-GPU validation runs) the parts
+                # are staged through host memory, as gloo requires
                 for gi, i in enumerate(sorted(outs)):
-                    torch.distributed.all_gather(gathered[gi], outs[i])
+                    src = outs[i] if gdev == "cuda" else outs[i].cpu()
+                    torch.distributed.all_gather(gathered[gi], src)
         alg_bytes_per_launch = S * STRIPE_BYTES * (K + M) // K
         metric = ("GiB/s EC mixed encode+3-erasure decode, ec(32,6) "
                   "64MiB stripes")
@@ -404,6 +412,28 @@ def main():
         t = torch.tensor([elapsed], device=dev, dtype=torch.float64)
         torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
         elapsed = float(t.item())
+
+    validated = None
+    if args.validate and args.op in ("decode", "mixed"):
+        # encode -> erase -> recover must reproduce the original bytes
+        # (bit-exact round trip; the GF math is deterministic)
+        for i in sorted(outs):
+            want_t = data[:, i, :] if i < K else parity[:, i - K, :]
+            if not torch.equal(outs[i], want_t):
+                log(f"[rank {rank}] VALIDATION FAILED: part {i} mismatch")
+                sys.exit(3)
+        if distributed and args.op == "mixed":
+            for gi, i in enumerate(sorted(outs)):
+                own = gathered[gi][rank]
+                src = outs[i] if own.is_cuda else outs[i].cpu()
+                if not torch.equal(own, src):
+                    log(f"[rank {rank}] VALIDATION FAILED: all-gather "
+                        f"slot {rank} part {i}")
+                    sys.exit(3)
+        validated = "recover round-trip bit-exact" + \
+            (", all-gather own-slot bit-exact"
+             if distributed and args.op == "mixed" else "")
+        log(f"[rank {rank}] validation OK: {validated}")
 
     step_ms = [a.elapsed_time(b) for a, b in ev]
     avg_step_ms = sum(step_ms) / len(step_ms)
@@ -465,9 +495,12 @@ def main():
             "stripe_bytes": STRIPE_BYTES,
             "stripes_per_gpu": S,
             "parallelism": f"independent stripe batches x{n_gpus}" +
-                           (" + RCCL all-gather of recovered parts"
+                           (f" + {args.backend} all-gather of recovered "
+                            f"parts"
                             if args.op == "mixed" and distributed else ""),
+            "backend": args.backend if distributed else None,
         },
+        "validated": validated,
         "roofline": roofline,
         "cpu_baseline": cpu_baseline,
     }

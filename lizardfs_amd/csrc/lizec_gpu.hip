@@ -3,18 +3,19 @@
  *
  * Replaces the reference's SIMD byte loop (galois_field_encode.cc:151-201,
  * AVX2 pshufb) with a CDNA4-native design:
- *  - GF(2^8) multiply via in-register nibble LUTs built from v_perm_b32
- *    (__builtin_amdgcn_perm): the 32-byte ISA-L coefficient table
- *    (tbl[i]=c*i, tbl[16+i]=c*(i<<4)) lives in 8 VGPRs per (dest,src) pair,
- *    re-read per 64-byte strip as a wave-uniform LDS broadcast;
+ *  - GF(2^8) multiply via in-register mixed-radix (3+3+2 bit field)
+ *    product LUTs selected with v_perm_b32 (__builtin_amdgcn_perm):
+ *    3 perms + 3 XORs per destination word, tables re-read per tile as
+ *    wave-uniform LDS broadcasts (ec_kernel.h gf_macc_mr);
  *  - 16-byte vectorized HBM loads/stores laid out so every wave instruction
  *    is a fully-coalesced 1 KiB access; full tiles run a branchless
  *    software-pipelined path (next source's strips prefetched while the
  *    current one is accumulated), ragged tails take a guarded path;
  *  - per-64KiB-block CRC32 (hddspacemgr.cc:1918 gate) with one wave per
- *    block: 64 lanes x 1 KiB segments, slicing-by-8 LDS tables, then a
- *    shfl-based combine tree folding segment CRCs with the same GF(2)
- *    "advance by N zero bytes" matrices as mycrc32_combine (crc.cc:153-224).
+ *    block via LDS-free carry-less folding (crc_fold.h) and a shfl-based
+ *    combine tree using the same GF(2) "advance by N zero bytes" matrices
+ *    as mycrc32_combine (crc.cc:153-224); slicing-table kernels remain
+ *    for odd block sizes.
  *
  * This file is the product compute path: it REQUIRES a GPU.  There is no
  * CPU fallback here by design (parity claims are void otherwise).
@@ -40,26 +41,33 @@
 #include "crc_fold.h"
 
 /* Product kernel configuration (chosen by the variant A/B harness,
- * bench_variants.hip; numbers in profiles/ROUND1.md):
+ * bench_variants.hip; numbers in profiles/ROUND1.md + ROUND2.md):
  *  - XCD-bijective block remap + non-temporal parity stores everywhere;
  *  - tile size per destination-group width D: D<=2 uses 4-chunk (16 KiB)
  *    tiles; D>=3 uses 2-chunk (8 KiB) tiles (fewer accumulator VGPRs ->
- *    4 waves/SIMD; ec(16,4) D4: 3953 -> 4113 GB/s, ec(32,6) D6: 3001). */
+ *    4 waves/SIMD; ec(16,4) D4: 3953 -> 4113 GB/s, ec(32,6) D6: 3001);
+ *  - mixed-radix (3+3+2) GF tables: 3 v_perms + 3 XORs per dest word
+ *    instead of the quarter-LUT's 4+4 — r2a/r2b A/B: ec(16,4) 4446 ->
+ *    5392 GB/s (+21%), ec(32,6) 3239 -> 3953 (+22%), ec(8,2) tie. */
 constexpr bool kECSwz = true;
 constexpr bool kECNtStore = true;
 constexpr bool kECNtLoad = true;   /* +3.6%: 5720 -> 5924 GB/s (profiles) */
-constexpr bool kECQuarterLut = true;  /* +4-9% everywhere; ec(8,2) 6197 GB/s */
 constexpr int ec_chunks_for(int d) { return d <= 4 ? 4 : 2; }
+constexpr int kECTblBytes = 32;    /* mixed-radix padded layout */
 
-/* repack one 32-byte ISA-L coefficient table into the 16-byte quarter-LUT
- * layout the QL kernel stages (see gf_macc_q in ec_kernel.h) */
-static void pack_quarter_lut(const uint8_t *t, uint8_t *q) {
-	for (int b = 0; b < 4; ++b) {
-		q[b] = t[b];
-		q[4 + b] = t[4 * b];
-		q[8 + b] = t[16 + b];
-		q[12 + b] = t[16 + 4 * b];
+/* repack one 32-byte ISA-L coefficient table into the padded 32-byte
+ * mixed-radix layout the MR kernel stages (see gf_macc_mr in ec_kernel.h):
+ * 8-entry product tables for bit fields [0:3) and [3:6), 4-entry for
+ * [6:8) — derived from the ISA-L lo/hi-nibble tables by GF linearity
+ * over disjoint bits. */
+static void pack_mixed_radix(const uint8_t *t, uint8_t *q) {
+	for (int v = 0; v < 8; ++v) {
+		q[v] = t[v];                                   /* c*v      */
+		q[8 + v] = t[(8 * v) & 15] ^ t[16 + (v >> 1)]; /* c*(v<<3) */
 	}
+	for (int v = 0; v < 4; ++v)
+		q[16 + v] = t[16 + (v << 2)];                  /* c*(v<<6) */
+	memset(q + 20, 0, 12);
 }
 
 /* ------------------------------------------------------------------ */
@@ -319,9 +327,9 @@ __global__ __launch_bounds__(kThreads) void scrub_chunks_kernel(
 		 * instantiation there (uint4 loads would be misaligned UB) */
 		uint32_t crc =
 		    (((uintptr_t)blockp & 15) == 0)
-		        ? crc_block_wave_fold<1, 2, true>(blockp, 65536u, 0u, T0,
+		        ? crc_block_wave_fold<1, 1, true>(blockp, 65536u, 0u, T0,
 		                                          mats, lane)
-		        : crc_block_wave_fold<1, 2, false>(blockp, 65536u, 0u, T0,
+		        : crc_block_wave_fold<1, 1, false>(blockp, 65536u, 0u, T0,
 		                                           mats, lane);
 		if (lane == 0) {
 			uint8_t *p = (uint8_t *)img + crc_offs[c] + crc_stride * b;
@@ -402,7 +410,7 @@ static int ctx_acquire(lizec_engine *e, hipStream_t s, size_t ptrs,
 		c.ev_recorded = false;
 	}
 	if (!c.d_gftbls)
-		LIZEC_CHECK(hipMalloc(&c.d_gftbls, (size_t)16 * 32 * 32));
+		LIZEC_CHECK(hipMalloc(&c.d_gftbls, (size_t)kECTblBytes * 32 * 32));
 	if (c.ptrs_cap < ptrs) {
 		size_t cap = c.ptrs_cap ? c.ptrs_cap : (1 << 16);
 		while (cap < ptrs) cap *= 2;
@@ -412,7 +420,7 @@ static int ctx_acquire(lizec_engine *e, hipStream_t s, size_t ptrs,
 		LIZEC_CHECK(hipMalloc(&c.d_ptrs, cap * sizeof(uint64_t)));
 		c.ptrs_cap = cap;
 	}
-	size_t want = staging + 16 * 32 * 32;
+	size_t want = staging + (size_t)kECTblBytes * 32 * 32;
 	if (c.h_cap < want) {
 		size_t cap = c.h_cap ? c.h_cap : (1 << 20);
 		while (cap < want) cap *= 2;
@@ -542,9 +550,9 @@ static void launch_ec(uint32_t part_len, int srcs, int dest_base,
 	uint32_t tiles_per_part = (part_len + tile_bytes - 1) / tile_bytes;
 	uint32_t total_tiles = tiles_per_part * nstripes;
 	uint32_t grid = total_tiles < 1048576u ? total_tiles : 1048576u;   /* exact grid (1 tile/block) measured +0.7% */
-	size_t lds = (size_t)D * srcs * (kECQuarterLut ? 16 : 32);
+	size_t lds = (size_t)D * srcs * kECTblBytes;
 	(void)tiles_per_part_unused;
-	hipLaunchKernelGGL(HIP_KERNEL_NAME(ec_encode_kernel<D, CH, kECSwz, kECNtStore, kECNtLoad, false, kECQuarterLut>),
+	hipLaunchKernelGGL(HIP_KERNEL_NAME(ec_encode_kernel<D, CH, kECSwz, kECNtStore, kECNtLoad, false, false, true>),
 	                   dim3(grid), dim3(kThreads), lds, s,
 	                   part_len, srcs, dest_base, d_tbls, d_src, d_dst,
 	                   dests_total, tiles_per_part, total_tiles);
@@ -614,12 +622,12 @@ extern "C" int lizec_ec_encode_batch(lizec_engine *e, uint64_t part_len,
 	/* stage tables + pointer arrays through this stream's pinned buffer
 	 * (ctx_acquire waited until any previous copies from it finished) */
 	uint8_t *tstage = c->h_staging;
-	uint64_t *pstage = (uint64_t *)(c->h_staging + 16 * 32 * 32);
+	uint64_t *pstage = (uint64_t *)(c->h_staging + (size_t)kECTblBytes * 32 * 32);
 	for (int i = 0; i < srcs * dests; ++i)
-		pack_quarter_lut(gftbls + (size_t)i * 32, tstage + (size_t)i * 16);
+		pack_mixed_radix(gftbls + (size_t)i * 32, tstage + (size_t)i * kECTblBytes);
 	memcpy(pstage, src_dptrs, nsrc * 8);
 	memcpy(pstage + nsrc, dst_dptrs, ndst * 8);
-	LIZEC_CHECK(hipMemcpyAsync(c->d_gftbls, tstage, (size_t)16 * srcs * dests,
+	LIZEC_CHECK(hipMemcpyAsync(c->d_gftbls, tstage, (size_t)kECTblBytes * srcs * dests,
 	                           hipMemcpyHostToDevice, s));
 	LIZEC_CHECK(hipMemcpyAsync(d_src, pstage, (nsrc + ndst) * 8,
 	                           hipMemcpyHostToDevice, s));
@@ -649,16 +657,16 @@ extern "C" int lizec_ec_plan_create(lizec_engine *e, uint64_t part_len,
 	p->num_stripes = num_stripes;
 	size_t nsrc = (size_t)num_stripes * srcs;
 	size_t ndst = (size_t)num_stripes * dests;
-	if (hipMalloc(&p->d_tbls, (size_t)16 * srcs * dests) != hipSuccess ||
+	if (hipMalloc(&p->d_tbls, (size_t)kECTblBytes * srcs * dests) != hipSuccess ||
 	    hipMalloc(&p->d_src, nsrc * 8) != hipSuccess ||
 	    hipMalloc(&p->d_dst, ndst * 8) != hipSuccess) {
 		lizec_ec_plan_destroy(p);
 		return LIZEC_ENOMEM;
 	}
-	uint8_t packed[16 * 32 * 32];
+	uint8_t packed[kECTblBytes * 32 * 32];
 	for (int i = 0; i < srcs * dests; ++i)
-		pack_quarter_lut(gftbls + (size_t)i * 32, packed + (size_t)i * 16);
-	if (hipMemcpy(p->d_tbls, packed, (size_t)16 * srcs * dests,
+		pack_mixed_radix(gftbls + (size_t)i * 32, packed + (size_t)i * kECTblBytes);
+	if (hipMemcpy(p->d_tbls, packed, (size_t)kECTblBytes * srcs * dests,
 	              hipMemcpyHostToDevice) != hipSuccess ||
 	    hipMemcpy(p->d_src, src_dptrs, nsrc * 8,
 	              hipMemcpyHostToDevice) != hipSuccess ||
@@ -703,11 +711,12 @@ extern "C" int lizec_crc32_batch(lizec_engine *e, const void *dev_buf,
 	const char *im = getenv("LIZEC_CRC_IMPL");
 	const char *fn = getenv("LIZEC_CRC_FOLD_NACC");
 	const char *nt = getenv("LIZEC_CRC_NT");
-	/* fold C=1 NACC=2 measured best (r2a: 4779 GB/s vs table 3474) */
+	/* fold C=1 NACC=1 measured best (r2b: 5112 GB/s = 0.64 of spec peak
+	 * vs table 3511; NT loads defeat the L1 line-burst reuse, -57%) */
 	int chains = ch ? atoi(ch) : 1;
 	int slice = sl ? atoi(sl) : 8;   /* slice-16 measured -16% within-box (profiles) */
 	bool fold = !(im && strcmp(im, "table") == 0);
-	int nacc = fn ? atoi(fn) : 2;
+	int nacc = fn ? atoi(fn) : 1;
 	bool ntld = nt && atoi(nt) != 0;
 	bool al16 = (((uintptr_t)dev_buf | block_len) & 15) == 0;
 	/* carry-less-folding path (default): block must split into C spans of
@@ -721,7 +730,7 @@ extern "C" int lizec_crc32_batch(lizec_engine *e, const void *dev_buf,
 	                   nblocks, seed, e->d_crc_const, dev_crcs_out)
 		if (!al16) {
 			if (chains == 2) LIZEC_LAUNCH_FOLD(2, 2, false, false);
-			else LIZEC_LAUNCH_FOLD(1, 2, false, false);
+			else LIZEC_LAUNCH_FOLD(1, 1, false, false);
 		} else if (chains == 2) {
 			if (nacc == 1) LIZEC_LAUNCH_FOLD(2, 1, true, false);
 			else LIZEC_LAUNCH_FOLD(2, 2, true, false);
@@ -782,7 +791,7 @@ extern "C" int lizec_scrub_batch_strided(
 	uint32_t *d_doffs = (uint32_t *)(d_ptrs + nchunks);
 	uint32_t *d_coffs = d_doffs + nchunks;
 	uint32_t *d_counts = d_coffs + nchunks;
-	uint8_t *st = c->h_staging + 16 * 32 * 32;
+	uint8_t *st = c->h_staging + (size_t)kECTblBytes * 32 * 32;
 	memcpy(st, chunk_dptrs, (size_t)nchunks * 8);
 	memcpy(st + (size_t)nchunks * 8, data_offs, (size_t)nchunks * 4);
 	memcpy(st + (size_t)nchunks * 12, crc_offs, (size_t)nchunks * 4);
@@ -928,12 +937,12 @@ extern "C" int lizec_replicate_run(lizec_engine *e, uint64_t part_len,
 	uint8_t *d_tbl = nullptr;
 	repl_slot sl[2];
 	int rc = LIZEC_OK;
-	uint8_t packed[16 * 32 * 32];
+	uint8_t packed[kECTblBytes * 32 * 32];
 	for (int i = 0; i < ic * oc; ++i)
-		pack_quarter_lut(gftbls + (size_t)i * 32, packed + (size_t)i * 16);
-	if (hipMalloc(&d_tbl, (size_t)16 * ic * oc) != hipSuccess)
+		pack_mixed_radix(gftbls + (size_t)i * 32, packed + (size_t)i * kECTblBytes);
+	if (hipMalloc(&d_tbl, (size_t)kECTblBytes * ic * oc) != hipSuccess)
 		return LIZEC_ENOMEM;
-	if (hipMemcpy(d_tbl, packed, (size_t)16 * ic * oc,
+	if (hipMemcpy(d_tbl, packed, (size_t)kECTblBytes * ic * oc,
 	              hipMemcpyHostToDevice) != hipSuccess) {
 		(void)hipFree(d_tbl);
 		return LIZEC_EHIP;

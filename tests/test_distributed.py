@@ -61,3 +61,71 @@ def test_weak_scaling_pattern_gloo_world2():
         assert ok, f"rank {rank} shard wrong"
         assert elapsed > 0
         assert total == 6
+
+
+def _mixed_worker(rank, world, port, q):
+    """bench.py --op mixed exchange structure on CPU: per-rank encode +
+    recover of erased data parts, then all_gather of the recovered parts
+    (SURVEY §8e — the one real exchange in the path); every rank verifies
+    BOTH gathered slots bit-exactly by recomputing each rank's
+    deterministic shard."""
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    torch.distributed.init_process_group("gloo", rank=rank, world_size=world)
+    import sys
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    sys.path.insert(0, os.path.join(repo, "oracle"))
+    import oracle
+
+    k, m, plen, stripes = 4, 2, 4096, 2
+    erased = (1, 3)
+
+    def shard(r):
+        rng = np.random.default_rng(200 + r)
+        return rng.integers(0, 256, (stripes, k, plen), np.uint8)
+
+    data = shard(rank)
+    parity = np.zeros((stripes, m, plen), np.uint8)
+    tbl, _, _ = oracle.rs_make_tables(k, m, (1 << k) - 1, (1 << k) - 1,
+                                      ((1 << m) - 1) << k)
+    oracle.encode_stripes(k, m, plen, stripes, tbl, data, parity)
+    # recover the erased data parts from the survivors
+    nparts = k + m
+    present = sum(1 << i for i in range(nparts) if i not in erased)
+    needed = sum(1 << i for i in erased)
+    rtbl, ic, oc = oracle.rs_make_tables(k, m, present, present, needed)
+    parts = np.concatenate([data, parity], axis=1)
+    srcs = np.ascontiguousarray(
+        parts[:, [i for i in range(nparts) if i not in erased][:ic]])
+    rec = np.zeros((stripes, oc, plen), np.uint8)
+    oracle.encode_stripes(ic, oc, plen, stripes, rtbl, srcs, rec)
+
+    gathered = [[torch.empty(stripes, plen, dtype=torch.uint8)
+                 for _ in range(world)] for _ in erased]
+    for gi in range(len(erased)):
+        torch.distributed.all_gather(
+            gathered[gi], torch.from_numpy(np.ascontiguousarray(rec[:, gi])))
+
+    ok = True
+    for r in range(world):
+        exp = shard(r)
+        for gi, i in enumerate(erased):
+            if not np.array_equal(gathered[gi][r].numpy(), exp[:, i]):
+                ok = False
+    torch.distributed.destroy_process_group()
+    q.put((rank, ok))
+
+
+def test_mixed_allgather_pattern_gloo_world2():
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_mixed_worker, args=(r, 2, 29519, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    results = [q.get(timeout=120) for _ in range(2)]
+    for p in procs:
+        p.join(timeout=60)
+        assert p.exitcode == 0
+    for rank, ok in results:
+        assert ok, f"rank {rank}: gathered recovered parts wrong"
