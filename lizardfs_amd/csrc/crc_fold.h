@@ -150,7 +150,7 @@ __device__ __forceinline__ uint4 crc_ld16(const uint8_t *p) {
  * a shfl tree with the advance matrices; lane 0 splices the C span CRCs.
  * Host guarantees block_len % (C * 64 * 16 * BV) == 0 (BV = 8, or 4 when
  * C = 4).  Returns the block CRC (lane 0's value is authoritative). */
-template <int C, int NACC, bool AL16 = true, bool NT = false>
+template <int C, int NACC, bool AL16 = true, bool NT = false, bool PF = false>
 __device__ uint32_t crc_block_wave_fold(const uint8_t *__restrict__ block,
                                         uint32_t block_len, uint32_t seed,
                                         const uint32_t *T0,
@@ -162,12 +162,22 @@ __device__ uint32_t crc_block_wave_fold(const uint8_t *__restrict__ block,
 	const uint8_t *base = block + (uint32_t)lane * seg;
 	uint32_t acc[C][NACC][4];
 	uint4 w[C][BV];
-	/* burst 0: seed the accumulators from the first NACC 16B groups */
+	auto ldburst = [&](uint4 (&buf)[C][BV], uint32_t i) {
 #pragma unroll
-	for (int c = 0; c < C; ++c)
+		for (int c = 0; c < C; ++c)
+#pragma unroll
+			for (int q = 0; q < BV; ++q)
+				buf[c][q] = crc_ld16<AL16, NT>(base + c * span + i + q * 16);
+	};
+	auto foldburst = [&](uint4 (&buf)[C][BV]) {
 #pragma unroll
 		for (int q = 0; q < BV; ++q)
-			w[c][q] = crc_ld16<AL16, NT>(base + c * span + q * 16);
+#pragma unroll
+			for (int c = 0; c < C; ++c)
+				crc_fold_step<NACC>(acc[c][q % NACC], buf[c][q]);
+	};
+	/* burst 0: seed the accumulators from the first NACC 16B groups */
+	ldburst(w, 0);
 #pragma unroll
 	for (int c = 0; c < C; ++c) {
 		const uint32_t raw0 =
@@ -183,17 +193,25 @@ __device__ uint32_t crc_block_wave_fold(const uint8_t *__restrict__ block,
 		for (int q = NACC; q < BV; ++q)
 			crc_fold_step<NACC>(acc[c][q % NACC], w[c][q]);
 	}
-	for (uint32_t i = 16 * BV; i < seg; i += 16 * BV) {
-#pragma unroll
-		for (int c = 0; c < C; ++c)
-#pragma unroll
-			for (int q = 0; q < BV; ++q)
-				w[c][q] = crc_ld16<AL16, NT>(base + c * span + i + q * 16);
-#pragma unroll
-		for (int q = 0; q < BV; ++q)
-#pragma unroll
-			for (int c = 0; c < C; ++c)
-				crc_fold_step<NACC>(acc[c][q % NACC], w[c][q]);
+	if (PF) {
+		/* software pipeline: the next burst's loads are in flight while
+		 * the current burst folds (two static buffers, 2 bursts/iter) */
+		uint4 w2[C][BV];
+		uint32_t i = 16 * BV;
+		if (i < seg) ldburst(w2, i);
+		for (; i < seg; i += 2 * 16 * BV) {
+			if (i + 16 * BV < seg) ldburst(w, i + 16 * BV);
+			foldburst(w2);
+			if (i + 16 * BV < seg) {
+				if (i + 2 * 16 * BV < seg) ldburst(w2, i + 2 * 16 * BV);
+				foldburst(w);
+			}
+		}
+	} else {
+		for (uint32_t i = 16 * BV; i < seg; i += 16 * BV) {
+			ldburst(w, i);
+			foldburst(w);
+		}
 	}
 	/* per-chain epilogue: merge the NACC interleaved accumulators with
 	 * 16-byte-distance folds, table-reduce, invert */

@@ -264,7 +264,7 @@ __global__ __launch_bounds__(kThreads) void crc32_blocks_kernel_multi(
  * LDS holds only the epilogue byte table T0 and the combine matrices. */
 constexpr int kCrcFoldLdsWords = 256 + kCrcMatCount * 32;
 
-template <int C, int NACC, bool AL16 = true, bool NT = false>
+template <int C, int NACC, bool AL16 = true, bool NT = false, bool PF = false>
 __global__ __launch_bounds__(kThreads) void crc32_blocks_kernel_fold(
     const uint8_t *__restrict__ buf, uint32_t block_len, uint64_t nblocks,
     uint32_t seed, const uint32_t *__restrict__ crc_const,
@@ -279,7 +279,7 @@ __global__ __launch_bounds__(kThreads) void crc32_blocks_kernel_fold(
 	const int lane = threadIdx.x & 63;
 	for (uint64_t blk = (uint64_t)blockIdx.x * 4 + wave; blk < nblocks;
 	     blk += (uint64_t)gridDim.x * 4) {
-		uint32_t crc = crc_block_wave_fold<C, NACC, AL16, NT>(
+		uint32_t crc = crc_block_wave_fold<C, NACC, AL16, NT, PF>(
 		    buf + blk * block_len, block_len, seed, stabs, stabs + 256, lane);
 		if (lane == 0) out[blk] = crc;
 	}
@@ -711,6 +711,7 @@ extern "C" int lizec_crc32_batch(lizec_engine *e, const void *dev_buf,
 	const char *im = getenv("LIZEC_CRC_IMPL");
 	const char *fn = getenv("LIZEC_CRC_FOLD_NACC");
 	const char *nt = getenv("LIZEC_CRC_NT");
+	const char *pf = getenv("LIZEC_CRC_PF");
 	/* fold C=1 NACC=1 measured best (r2b: 5112 GB/s = 0.64 of spec peak
 	 * vs table 3511; NT loads defeat the L1 line-burst reuse, -57%) */
 	int chains = ch ? atoi(ch) : 1;
@@ -718,6 +719,7 @@ extern "C" int lizec_crc32_batch(lizec_engine *e, const void *dev_buf,
 	bool fold = !(im && strcmp(im, "table") == 0);
 	int nacc = fn ? atoi(fn) : 1;
 	bool ntld = nt && atoi(nt) != 0;
+	bool pfld = pf && atoi(pf) != 0;
 	bool al16 = (((uintptr_t)dev_buf | block_len) & 15) == 0;
 	/* carry-less-folding path (default): block must split into C spans of
 	 * whole 64-lane x BV*16-byte bursts (BV=8) */
@@ -734,6 +736,10 @@ extern "C" int lizec_crc32_batch(lizec_engine *e, const void *dev_buf,
 		} else if (chains == 2) {
 			if (nacc == 1) LIZEC_LAUNCH_FOLD(2, 1, true, false);
 			else LIZEC_LAUNCH_FOLD(2, 2, true, false);
+		} else if (pfld) {
+			hipLaunchKernelGGL(HIP_KERNEL_NAME(crc32_blocks_kernel_fold<1, 1, true, false, true>),
+			                   dim3(grid), dim3(kThreads), 0, s, b, block_len,
+			                   nblocks, seed, e->d_crc_const, dev_crcs_out);
 		} else if (ntld) {
 			if (nacc == 1) LIZEC_LAUNCH_FOLD(1, 1, true, true);
 			else if (nacc == 4) LIZEC_LAUNCH_FOLD(1, 4, true, true);
