@@ -80,6 +80,12 @@ def _load_libref():
                                  ctypes.c_uint64, pp, ctypes.c_size_t]
     L.ref_mycrc32.restype = ctypes.c_uint32
     L.ref_mycrc32.argtypes = [ctypes.c_uint32, u8p, ctypes.c_uint32]
+    try:
+        L.ref_mycrc32_blocks.argtypes = [u8p, ctypes.c_uint64,
+                                         ctypes.c_uint32,
+                                         ctypes.POINTER(ctypes.c_uint32)]
+    except AttributeError:
+        pass   # older prebuilt _ref: per-block calls instead
     L.ref_mycrc32_init()
     return L
 
@@ -173,17 +179,26 @@ def cpu_baseline_leg(op, k, m, erased, target_seconds=6.0, max_reps=64):
             legs.append(lambda: list(pool.map(rec_one, tasks)))
         if op in ("crc", "encode_crc"):
             u8p = ctypes.POINTER(ctypes.c_uint8)
+            u32p = ctypes.POINTER(ctypes.c_uint32)
             rows = [data[s].reshape(-1) for s in range(n)]
+            crcs = [np.empty(rows[0].size // 65536, np.uint32)
+                    for _ in range(n)]
             crc_seg = max(65536, (rows[0].size // segs) & ~65535)
             crc_tasks = [(s, off, min(crc_seg, rows[0].size - off))
                          for s in range(n)
                          for off in range(0, rows[0].size, crc_seg)]
+            blocks_fn = getattr(ref, "ref_mycrc32_blocks", None)
 
             def crc_one(t):
                 s, off, ln = t
                 base = rows[s].ctypes.data + off
-                for o in range(0, ln, 65536):
-                    ref.ref_mycrc32(0, ctypes.cast(base + o, u8p), 65536)
+                if blocks_fn is not None:
+                    blocks_fn(ctypes.cast(base, u8p), ln // 65536, 65536,
+                              ctypes.cast(crcs[s].ctypes.data +
+                                          (off // 65536) * 4, u32p))
+                else:
+                    for o in range(0, ln, 65536):
+                        ref.ref_mycrc32(0, ctypes.cast(base + o, u8p), 65536)
             legs.append(lambda: list(pool.map(crc_one, crc_tasks)))
         kind = "reference"
         how = (f"reference AVX2 (oracle/_ref/libref.so), {cores} threads "

@@ -53,6 +53,14 @@ int ref_rs_recover(int k, int m, const uint8_t **fragments,
 uint32_t ref_mycrc32(uint32_t crc, const uint8_t *block, uint32_t leng) {
 	return mycrc32(crc, block, leng);
 }
+
+/* Per-block CRC loop in C so threaded bench callers are not bound by the
+ * Python GIL between 64 KiB calls (one foreign call covers many blocks). */
+void ref_mycrc32_blocks(const uint8_t *buf, uint64_t nblocks,
+                        uint32_t block_len, uint32_t *out) {
+	for (uint64_t b = 0; b < nblocks; ++b)
+		out[b] = mycrc32(0, buf + b * (uint64_t)block_len, block_len);
+}
 uint32_t ref_mycrc32_combine(uint32_t crc1, uint32_t crc2, uint32_t leng2) {
 	return mycrc32_combine(crc1, crc2, leng2);
 }
