@@ -154,6 +154,8 @@ int main(int argc, char **argv) {
 	    {"D4_CH4_mr    ", launch_var<4, 4, true, true, true, false, false, true>, 4, 4, 1048576, 2},
 	    {"D4_CH2_mr    ", launch_var<4, 2, true, true, true, false, false, true>, 2, 4, 1048576, 2},
 	    {"D6_CH2_mr    ", launch_var<6, 2, true, true, true, false, false, true>, 2, 6, 1048576, 2},
+	    {"D6_CH1_mr    ", launch_var<6, 1, true, true, true, false, false, true>, 1, 6, 1048576, 2},
+	    {"D3_CH4_mr    ", launch_var<3, 4, true, true, true, false, false, true>, 4, 3, 1048576, 2},
 	    {"D6_CH4_mr    ", launch_var<6, 4, true, true, true, false, false, true>, 4, 6, 1048576, 2},
 	    {"D8_CH2_mr    ", launch_var<8, 2, true, true, true, false, false, true>, 2, 8, 1048576, 2},
 	    {"D1_CH4_swz_nt ", launch_var<1, 4, true, true>, 4, 1, 262144},

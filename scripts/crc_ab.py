@@ -17,19 +17,16 @@ import torch  # noqa: E402
 
 from lizardfs_amd import crc as lcrc  # noqa: E402
 
-GIB = 16  # device buffer size
+GIB = int(sys.argv[1]) if len(sys.argv) > 1 else 16  # device buffer size
 BLOCK = 65536
 REPS = 10
 
 CFGS = [
-    # name, impl, chains, nacc, nt
-    ("table_c2_s8", "table", "2", "", "0"),
-    ("fold_c1_n1 ", "fold", "1", "1", "0"),
-    ("fold_c1_n2 ", "fold", "1", "2", "0"),
-    ("fold_c1_n4 ", "fold", "1", "4", "0"),
-    ("fold_c1_n1nt", "fold", "1", "1", "1"),
-    ("fold_c1_n2nt", "fold", "1", "2", "1"),
-    ("fold_c2_n2 ", "fold", "2", "2", "0"),
+    # name, impl, chains, nacc, nt, pf
+    ("table_c2_s8", "table", "2", "", "0", "0"),
+    ("fold_c1_n1 ", "fold", "1", "1", "0", "0"),
+    ("fold_c1_n1pf", "fold", "1", "1", "0", "1"),
+    ("fold_c1_n4 ", "fold", "1", "4", "0", "0"),
 ]
 
 
@@ -44,11 +41,13 @@ def main():
     ref = None
     e0 = torch.cuda.Event(enable_timing=True)
     e1 = torch.cuda.Event(enable_timing=True)
-    for name, impl, chains, nacc, nt in CFGS:
+    print(f"crc_ab buffer {GIB} GiB", flush=True)
+    for name, impl, chains, nacc, nt, pf in CFGS:
         os.environ["LIZEC_CRC_IMPL"] = impl
         os.environ["LIZEC_CRC_CHAINS"] = chains
         os.environ["LIZEC_CRC_FOLD_NACC"] = nacc or "2"
         os.environ["LIZEC_CRC_NT"] = nt
+        os.environ["LIZEC_CRC_PF"] = pf
         out.zero_()
         torch.cuda.synchronize()
         for _ in range(3):
@@ -74,6 +73,7 @@ def main():
     os.environ["LIZEC_CRC_IMPL"] = "fold"
     os.environ["LIZEC_CRC_CHAINS"] = "1"
     os.environ["LIZEC_CRC_NT"] = "0"
+    os.environ["LIZEC_CRC_PF"] = "0"
     small = buf[:3 * 1024 * 1024]
     c1 = lcrc.crc32_blocks(small, 3072, seed=0xDEADBEEF)
     os.environ["LIZEC_CRC_IMPL"] = "table"
