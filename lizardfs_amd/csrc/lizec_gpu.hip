@@ -922,9 +922,10 @@ extern "C" int lizec_replicate_run(lizec_engine *e, uint64_t part_len,
 	const uint64_t img_bytes = header_size + part_len;
 	int B = sub_batch;
 	if (B <= 0) {
-		/* ~1 GiB of device staging per slot */
+		/* ~2 GiB of device staging per slot (sub_batch 16 at ec(8,2)
+		 * measured ~10% over 12; profiles/ROUND2.md) */
 		uint64_t per_chunk = (uint64_t)ic * part_len + oc * img_bytes;
-		B = (int)(((uint64_t)1 << 30) / per_chunk);
+		B = (int)(((uint64_t)1 << 31) / per_chunk);
 		if (B < 1) B = 1;
 		if (B > 64) B = 64;
 	}
