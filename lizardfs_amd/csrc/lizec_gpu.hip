@@ -327,10 +327,10 @@ __global__ __launch_bounds__(kThreads) void scrub_chunks_kernel(
 		 * instantiation there (uint4 loads would be misaligned UB) */
 		uint32_t crc =
 		    (((uintptr_t)blockp & 15) == 0)
-		        ? crc_block_wave_fold<1, 1, true>(blockp, 65536u, 0u, T0,
-		                                          mats, lane)
-		        : crc_block_wave_fold<1, 1, false>(blockp, 65536u, 0u, T0,
-		                                           mats, lane);
+		        ? crc_block_wave_fold<1, 1, true, false, true>(
+		              blockp, 65536u, 0u, T0, mats, lane)
+		        : crc_block_wave_fold<1, 1, false, false, true>(
+		              blockp, 65536u, 0u, T0, mats, lane);
 		if (lane == 0) {
 			uint8_t *p = (uint8_t *)img + crc_offs[c] + crc_stride * b;
 			if (WRITE) {
@@ -719,7 +719,8 @@ extern "C" int lizec_crc32_batch(lizec_engine *e, const void *dev_buf,
 	bool fold = !(im && strcmp(im, "table") == 0);
 	int nacc = fn ? atoi(fn) : 1;
 	bool ntld = nt && atoi(nt) != 0;
-	bool pfld = pf && atoi(pf) != 0;
+	/* burst prefetch default-on (r2d: +4-7%, 5147 GB/s at 64 GiB) */
+	bool pfld = pf ? atoi(pf) != 0 : true;
 	bool al16 = (((uintptr_t)dev_buf | block_len) & 15) == 0;
 	/* carry-less-folding path (default): block must split into C spans of
 	 * whole 64-lane x BV*16-byte bursts (BV=8) */
