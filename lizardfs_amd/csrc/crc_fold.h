@@ -150,12 +150,13 @@ __device__ __forceinline__ uint4 crc_ld16(const uint8_t *p) {
  * a shfl tree with the advance matrices; lane 0 splices the C span CRCs.
  * Host guarantees block_len % (C * 64 * 16 * BV) == 0 (BV = 8, or 4 when
  * C = 4).  Returns the block CRC (lane 0's value is authoritative). */
-template <int C, int NACC, bool AL16 = true, bool NT = false, bool PF = false>
+template <int C, int NACC, bool AL16 = true, bool NT = false,
+          bool PF = false, int BVO = 0>
 __device__ uint32_t crc_block_wave_fold(const uint8_t *__restrict__ block,
                                         uint32_t block_len, uint32_t seed,
                                         const uint32_t *T0,
                                         const uint32_t *mats, int lane) {
-	constexpr int BV = C >= 4 ? 4 : 8;
+	constexpr int BV = BVO ? BVO : (C >= 4 ? 4 : 8);
 	static_assert(BV % NACC == 0, "burst must cover whole interleave groups");
 	const uint32_t span = block_len / C;
 	const uint32_t seg = span >> 6; /* bytes per lane, % (16*BV) == 0 */

@@ -264,7 +264,8 @@ __global__ __launch_bounds__(kThreads) void crc32_blocks_kernel_multi(
  * LDS holds only the epilogue byte table T0 and the combine matrices. */
 constexpr int kCrcFoldLdsWords = 256 + kCrcMatCount * 32;
 
-template <int C, int NACC, bool AL16 = true, bool NT = false, bool PF = false>
+template <int C, int NACC, bool AL16 = true, bool NT = false,
+          bool PF = false, int BVO = 0>
 __global__ __launch_bounds__(kThreads) void crc32_blocks_kernel_fold(
     const uint8_t *__restrict__ buf, uint32_t block_len, uint64_t nblocks,
     uint32_t seed, const uint32_t *__restrict__ crc_const,
@@ -279,7 +280,7 @@ __global__ __launch_bounds__(kThreads) void crc32_blocks_kernel_fold(
 	const int lane = threadIdx.x & 63;
 	for (uint64_t blk = (uint64_t)blockIdx.x * 4 + wave; blk < nblocks;
 	     blk += (uint64_t)gridDim.x * 4) {
-		uint32_t crc = crc_block_wave_fold<C, NACC, AL16, NT, PF>(
+		uint32_t crc = crc_block_wave_fold<C, NACC, AL16, NT, PF, BVO>(
 		    buf + blk * block_len, block_len, seed, stabs, stabs + 256, lane);
 		if (lane == 0) out[blk] = crc;
 	}
@@ -738,9 +739,17 @@ extern "C" int lizec_crc32_batch(lizec_engine *e, const void *dev_buf,
 			if (nacc == 1) LIZEC_LAUNCH_FOLD(2, 1, true, false);
 			else LIZEC_LAUNCH_FOLD(2, 2, true, false);
 		} else if (pfld) {
-			hipLaunchKernelGGL(HIP_KERNEL_NAME(crc32_blocks_kernel_fold<1, 1, true, false, true>),
-			                   dim3(grid), dim3(kThreads), 0, s, b, block_len,
-			                   nblocks, seed, e->d_crc_const, dev_crcs_out);
+			const char *bv = getenv("LIZEC_CRC_BV");
+			if (bv && atoi(bv) == 4)
+				hipLaunchKernelGGL(HIP_KERNEL_NAME(crc32_blocks_kernel_fold<1, 1, true, false, true, 4>),
+				                   dim3(grid), dim3(kThreads), 0, s, b,
+				                   block_len, nblocks, seed, e->d_crc_const,
+				                   dev_crcs_out);
+			else
+				hipLaunchKernelGGL(HIP_KERNEL_NAME(crc32_blocks_kernel_fold<1, 1, true, false, true>),
+				                   dim3(grid), dim3(kThreads), 0, s, b,
+				                   block_len, nblocks, seed, e->d_crc_const,
+				                   dev_crcs_out);
 		} else if (ntld) {
 			if (nacc == 1) LIZEC_LAUNCH_FOLD(1, 1, true, true);
 			else if (nacc == 4) LIZEC_LAUNCH_FOLD(1, 4, true, true);
