@@ -22,11 +22,11 @@ BLOCK = 65536
 REPS = 10
 
 CFGS = [
-    # name, impl, chains, nacc, nt, pf
-    ("table_c2_s8", "table", "2", "", "0", "0"),
-    ("fold_c1_n1 ", "fold", "1", "1", "0", "0"),
-    ("fold_c1_n1pf", "fold", "1", "1", "0", "1"),
-    ("fold_c1_n4 ", "fold", "1", "4", "0", "0"),
+    # name, impl, chains, nacc, nt, pf, bv
+    ("table_c2_s8", "table", "2", "", "0", "0", "0"),
+    ("fold_c1_n1 ", "fold", "1", "1", "0", "0", "0"),
+    ("fold_pf_bv8", "fold", "1", "1", "0", "1", "0"),
+    ("fold_pf_bv4", "fold", "1", "1", "0", "1", "4"),
 ]
 
 
@@ -42,12 +42,13 @@ def main():
     e0 = torch.cuda.Event(enable_timing=True)
     e1 = torch.cuda.Event(enable_timing=True)
     print(f"crc_ab buffer {GIB} GiB", flush=True)
-    for name, impl, chains, nacc, nt, pf in CFGS:
+    for name, impl, chains, nacc, nt, pf, bv in CFGS:
         os.environ["LIZEC_CRC_IMPL"] = impl
         os.environ["LIZEC_CRC_CHAINS"] = chains
         os.environ["LIZEC_CRC_FOLD_NACC"] = nacc or "2"
         os.environ["LIZEC_CRC_NT"] = nt
         os.environ["LIZEC_CRC_PF"] = pf
+        os.environ["LIZEC_CRC_BV"] = bv
         out.zero_()
         torch.cuda.synchronize()
         for _ in range(3):
