@@ -35,7 +35,11 @@ def test_exports_complete():
                 "lizec_chunk_part_index", "lizec_chunk_part_length",
                 "lizec_gpu_count", "lizec_engine_create",
                 "lizec_engine_destroy", "lizec_engine_sync",
-                "lizec_ec_encode_batch", "lizec_crc32_batch"]:
+                "lizec_ec_encode_batch", "lizec_crc32_batch",
+                "lizec_ec_plan_create", "lizec_ec_plan_run",
+                "lizec_ec_plan_destroy", "lizec_scrub_batch",
+                "lizec_scrub_batch_strided", "lizec_host_alloc",
+                "lizec_host_free", "lizec_replicate_run"]:
         assert getattr(lib, sym, None) is not None, sym
 
 
