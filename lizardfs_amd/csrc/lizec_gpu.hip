@@ -720,8 +720,11 @@ extern "C" int lizec_crc32_batch(lizec_engine *e, const void *dev_buf,
 	bool fold = !(im && strcmp(im, "table") == 0);
 	int nacc = fn ? atoi(fn) : 1;
 	bool ntld = nt && atoi(nt) != 0;
-	/* burst prefetch default-on (r2d: +4-7%, 5147 GB/s at 64 GiB) */
-	bool pfld = pf ? atoi(pf) != 0 : true;
+	/* burst prefetch measured box-DEPENDENT: +7% on one box (r2d,
+	 * 5147 GB/s) but -20% on another (r2e, 4154 vs 5197 no-PF) — its
+	 * 2-waves/SIMD occupancy is fragile, the 5-wave no-PF shape is
+	 * robust across boxes.  Default off; env opt-in for A/B. */
+	bool pfld = pf && atoi(pf) != 0;
 	bool al16 = (((uintptr_t)dev_buf | block_len) & 15) == 0;
 	/* carry-less-folding path (default): block must split into C spans of
 	 * whole 64-lane x BV*16-byte bursts (BV=8) */
