@@ -328,10 +328,10 @@ __global__ __launch_bounds__(kThreads) void scrub_chunks_kernel(
 		 * instantiation there (uint4 loads would be misaligned UB) */
 		uint32_t crc =
 		    (((uintptr_t)blockp & 15) == 0)
-		        ? crc_block_wave_fold<1, 1, true, false, true>(
-		              blockp, 65536u, 0u, T0, mats, lane)
-		        : crc_block_wave_fold<1, 1, false, false, true>(
-		              blockp, 65536u, 0u, T0, mats, lane);
+		        ? crc_block_wave_fold<1, 1, true>(blockp, 65536u, 0u, T0,
+		                                          mats, lane)
+		        : crc_block_wave_fold<1, 1, false>(blockp, 65536u, 0u, T0,
+		                                           mats, lane);
 		if (lane == 0) {
 			uint8_t *p = (uint8_t *)img + crc_offs[c] + crc_stride * b;
 			if (WRITE) {
