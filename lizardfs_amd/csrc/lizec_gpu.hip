@@ -758,7 +758,13 @@ extern "C" int lizec_crc32_batch(lizec_engine *e, const void *dev_buf,
 			else if (nacc == 4) LIZEC_LAUNCH_FOLD(1, 4, true, true);
 			else LIZEC_LAUNCH_FOLD(1, 2, true, true);
 		} else {
-			if (nacc == 1) LIZEC_LAUNCH_FOLD(1, 1, true, false);
+			const char *bv = getenv("LIZEC_CRC_BV");
+			if (bv && atoi(bv) == 4)
+				hipLaunchKernelGGL(HIP_KERNEL_NAME(crc32_blocks_kernel_fold<1, 1, true, false, false, 4>),
+				                   dim3(grid), dim3(kThreads), 0, s, b,
+				                   block_len, nblocks, seed, e->d_crc_const,
+				                   dev_crcs_out);
+			else if (nacc == 1) LIZEC_LAUNCH_FOLD(1, 1, true, false);
 			else if (nacc == 4) LIZEC_LAUNCH_FOLD(1, 4, true, false);
 			else LIZEC_LAUNCH_FOLD(1, 2, true, false);
 		}

@@ -27,6 +27,7 @@ CFGS = [
     ("fold_c1_n1 ", "fold", "1", "1", "0", "0", "0"),
     ("fold_pf_bv8", "fold", "1", "1", "0", "1", "0"),
     ("fold_pf_bv4", "fold", "1", "1", "0", "1", "4"),
+    ("fold_bv4   ", "fold", "1", "1", "0", "0", "4"),
 ]
 
 

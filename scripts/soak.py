@@ -2,8 +2,11 @@
 """Stability soak for the adopted kernels: continuous encode + 2-erasure
 recover + per-block CRC on one batch, output bit-compared against the
 first iteration every 50 rounds.  Usage: soak.py [seconds]"""
+import os
 import sys
 import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import torch
 
