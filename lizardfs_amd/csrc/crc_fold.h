@@ -11,9 +11,10 @@
  *
  * with clmul against the compile-time constants emitted by
  * tools/gen_crc_fold.py (solved from the mycrc32 semantics, crc.cc:113-151,
- * and minimized to ~13 set bits each over the 32-dim solution coset).  The
- * constant tests unroll away, so one fold is ~26 shifted-XOR terms of
- * straight-line VALU code — no LDS in the hot loop, no serial table chain.
+ * and minimized to 8-9 set bits each by information-set decoding over the
+ * 32-dim solution coset).  The constant tests unroll away, so one fold is
+ * ~17 shifted-XOR terms of straight-line VALU code — no LDS in the hot
+ * loop, no serial table chain.
  * LDS is used only in the epilogue: a single 256-entry byte table reduces
  * the final 16-byte accumulator (16 lookups per lane segment), and the
  * mycrc32_combine advance matrices (crc.cc:153-224) drive the cross-lane
