@@ -4,6 +4,8 @@ Covers the full supported range (k in [2,32], m in [1,32], random NULL
 zero-parts, random erasures incl. mixed data/parity, random part lengths
 incl. ragged tile tails) beyond the fixed golden cases.
 """
+import os
+
 import numpy as np
 import pytest
 
@@ -11,13 +13,16 @@ import oracle
 
 pytestmark = pytest.mark.gpu
 
+SEED = int(os.environ.get("LIZEC_SWEEP_SEED", "31337"))
+TRIALS = int(os.environ.get("LIZEC_SWEEP_TRIALS", "24"))
+
 
 def test_random_config_sweep():
     import torch
     from lizardfs_amd.ec import ReedSolomon
 
-    rng = np.random.default_rng(31337)
-    for trial in range(24):
+    rng = np.random.default_rng(SEED)
+    for trial in range(TRIALS):
         k = int(rng.integers(2, 33))
         m = int(rng.integers(1, 33))
         nparts = k + m
@@ -97,3 +102,23 @@ def test_full_size_roundtrip_wide_configs():
         for i in erase:
             expect = data[:, i] if i < k else parity[:, i - k]
             assert torch.equal(rec[i], expect), (k, m, i)
+
+
+def test_random_crc_block_sweep():
+    """Random block sizes/seeds across the CRC dispatch boundaries
+    (fold fast path % 8 KiB, table fast path % 16 KiB, generic % 1 KiB),
+    bit-exact vs the oracle."""
+    import torch
+    from lizardfs_amd import crc as crc_mod
+
+    rng = np.random.default_rng(SEED + 1)
+    for trial in range(max(8, TRIALS // 2)):
+        blen = int(rng.integers(1, 257)) * 1024
+        nblocks = int(rng.integers(1, 9))
+        seed = int(rng.integers(0, 2**32)) if trial % 3 else 0
+        buf = rng.integers(0, 256, nblocks * blen, np.uint8)
+        got = crc_mod.crc32_blocks(
+            torch.from_numpy(buf).cuda(), blen,
+            seed=seed).cpu().numpy().view(np.uint32)
+        exp = oracle.crc32_blocks(buf, blen, seed=seed)
+        assert np.array_equal(got, exp), (trial, blen, nblocks, hex(seed))
