@@ -26,6 +26,7 @@
 #include <cstdio>
 #include <cstring>
 #include <cstdlib>
+#include <atomic>
 #include <mutex>
 #include <unordered_map>
 
@@ -376,6 +377,10 @@ struct lizec_engine {
 	uint32_t *d_crc_const;     /* kCrcConstWords */
 	std::mutex mu;             /* guards ctxs */
 	std::unordered_map<void *, lizec_stream_ctx> ctxs;
+	/* CRC fold shape, autotuned on the first large batch: -1 undecided,
+	 * 0 = no-prefetch (5 waves/SIMD, robust), 1 = burst-prefetch
+	 * (2 waves; +6-7% on most boxes, -20% on some — profiles/ROUND2.md) */
+	std::atomic<int> crc_shape{-1};
 };
 
 static void ctx_free(lizec_stream_ctx &c) {
@@ -759,14 +764,61 @@ extern "C" int lizec_crc32_batch(lizec_engine *e, const void *dev_buf,
 			else LIZEC_LAUNCH_FOLD(1, 2, true, true);
 		} else {
 			const char *bv = getenv("LIZEC_CRC_BV");
-			if (bv && atoi(bv) == 4)
+			const char *at = getenv("LIZEC_CRC_AUTOTUNE");
+			bool tune = !(at && atoi(at) == 0);
+			if (bv && atoi(bv) == 4) {
 				hipLaunchKernelGGL(HIP_KERNEL_NAME(crc32_blocks_kernel_fold<1, 1, true, false, false, 4>),
 				                   dim3(grid), dim3(kThreads), 0, s, b,
 				                   block_len, nblocks, seed, e->d_crc_const,
 				                   dev_crcs_out);
-			else if (nacc == 1) LIZEC_LAUNCH_FOLD(1, 1, true, false);
-			else if (nacc == 4) LIZEC_LAUNCH_FOLD(1, 4, true, false);
-			else LIZEC_LAUNCH_FOLD(1, 2, true, false);
+			} else if (nacc == 4) {
+				LIZEC_LAUNCH_FOLD(1, 4, true, false);
+			} else if (nacc == 2) {
+				LIZEC_LAUNCH_FOLD(1, 2, true, false);
+			} else {
+				int shape = e->crc_shape.load(std::memory_order_relaxed);
+				if (shape < 0 && tune && nblocks >= 4096) {
+					/* Autotune once per engine: time both shapes on this
+					 * batch.  Both kernels emit the correct CRCs, so the
+					 * probe costs one extra pass and the call stays
+					 * correct even if timing fails. */
+					hipEvent_t ev[3];
+					int made = 0;
+					for (; made < 3; ++made)
+						if (hipEventCreate(&ev[made]) != hipSuccess) break;
+					if (made == 3) {
+						(void)hipEventRecord(ev[0], s);
+						LIZEC_LAUNCH_FOLD(1, 1, true, false);
+						(void)hipEventRecord(ev[1], s);
+						hipLaunchKernelGGL(HIP_KERNEL_NAME(crc32_blocks_kernel_fold<1, 1, true, false, true>),
+						                   dim3(grid), dim3(kThreads), 0, s,
+						                   b, block_len, nblocks, seed,
+						                   e->d_crc_const, dev_crcs_out);
+						(void)hipEventRecord(ev[2], s);
+						if (hipEventSynchronize(ev[2]) == hipSuccess) {
+							float t0 = 0.f, t1 = 0.f;
+							(void)hipEventElapsedTime(&t0, ev[0], ev[1]);
+							(void)hipEventElapsedTime(&t1, ev[1], ev[2]);
+							shape = (t1 > 0.f && t1 < t0) ? 1 : 0;
+							e->crc_shape.store(shape,
+							                   std::memory_order_relaxed);
+						}
+					} else {
+						LIZEC_LAUNCH_FOLD(1, 1, true, false);
+					}
+					for (int i = 0; i < made; ++i)
+						(void)hipEventDestroy(ev[i]);
+					LIZEC_CHECK(hipGetLastError());
+					return LIZEC_OK;
+				}
+				if (shape == 1)
+					hipLaunchKernelGGL(HIP_KERNEL_NAME(crc32_blocks_kernel_fold<1, 1, true, false, true>),
+					                   dim3(grid), dim3(kThreads), 0, s, b,
+					                   block_len, nblocks, seed,
+					                   e->d_crc_const, dev_crcs_out);
+				else
+					LIZEC_LAUNCH_FOLD(1, 1, true, false);
+			}
 		}
 #undef LIZEC_LAUNCH_FOLD
 		LIZEC_CHECK(hipGetLastError());
