@@ -213,3 +213,38 @@ def test_native_library_is_loaded():
     assert os.path.exists(so)
     assert L.lib()._name == so
     assert L.lib().lizec_gpu_count() >= 1
+
+
+def test_crc_fold_shapes_agree(crc_mod):
+    """Every fold shape the autotuner or env hooks can pick (no-PF,
+    burst-prefetch, BV=4 variants) is bit-identical to the pinned default
+    on a batch large enough to trigger autotuning (>= 4096 blocks)."""
+    import torch
+    g = torch.Generator(device="cuda").manual_seed(5)
+    buf = torch.randint(0, 256, (4096 * 65536,), dtype=torch.uint8,
+                        device="cuda", generator=g)
+    saved = {k: os.environ.get(k) for k in
+             ("LIZEC_CRC_AUTOTUNE", "LIZEC_CRC_PF", "LIZEC_CRC_BV")}
+    try:
+        os.environ["LIZEC_CRC_AUTOTUNE"] = "0"
+        os.environ["LIZEC_CRC_PF"] = "0"
+        os.environ["LIZEC_CRC_BV"] = "0"
+        ref = crc_mod.crc32_blocks(buf, 65536).clone()
+        torch.cuda.synchronize()
+        for env in ({"LIZEC_CRC_PF": "1"},
+                    {"LIZEC_CRC_PF": "1", "LIZEC_CRC_BV": "4"},
+                    {"LIZEC_CRC_BV": "4"},
+                    {"LIZEC_CRC_AUTOTUNE": "1"}):
+            os.environ.update(env)
+            got = crc_mod.crc32_blocks(buf, 65536)
+            torch.cuda.synchronize()
+            assert torch.equal(got, ref), env
+            os.environ["LIZEC_CRC_PF"] = "0"
+            os.environ["LIZEC_CRC_BV"] = "0"
+            os.environ["LIZEC_CRC_AUTOTUNE"] = "0"
+    finally:
+        for k, v in saved.items():
+            if v is None:
+                os.environ.pop(k, None)
+            else:
+                os.environ[k] = v
