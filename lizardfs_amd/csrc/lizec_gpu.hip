@@ -766,15 +766,27 @@ extern "C" int lizec_crc32_batch(lizec_engine *e, const void *dev_buf,
 			const char *bv = getenv("LIZEC_CRC_BV");
 			const char *at = getenv("LIZEC_CRC_AUTOTUNE");
 			bool tune = !(at && atoi(at) == 0);
-			if (bv && atoi(bv) == 4) {
+			if (bv && atoi(bv) == 4 && nacc == 1) {
 				hipLaunchKernelGGL(HIP_KERNEL_NAME(crc32_blocks_kernel_fold<1, 1, true, false, false, 4>),
 				                   dim3(grid), dim3(kThreads), 0, s, b,
 				                   block_len, nblocks, seed, e->d_crc_const,
 				                   dev_crcs_out);
 			} else if (nacc == 4) {
-				LIZEC_LAUNCH_FOLD(1, 4, true, false);
+				if (bv && atoi(bv) == 4)
+					hipLaunchKernelGGL(HIP_KERNEL_NAME(crc32_blocks_kernel_fold<1, 4, true, false, false, 4>),
+					                   dim3(grid), dim3(kThreads), 0, s, b,
+					                   block_len, nblocks, seed,
+					                   e->d_crc_const, dev_crcs_out);
+				else
+					LIZEC_LAUNCH_FOLD(1, 4, true, false);
 			} else if (nacc == 2) {
-				LIZEC_LAUNCH_FOLD(1, 2, true, false);
+				if (bv && atoi(bv) == 4)
+					hipLaunchKernelGGL(HIP_KERNEL_NAME(crc32_blocks_kernel_fold<1, 2, true, false, false, 4>),
+					                   dim3(grid), dim3(kThreads), 0, s, b,
+					                   block_len, nblocks, seed,
+					                   e->d_crc_const, dev_crcs_out);
+				else
+					LIZEC_LAUNCH_FOLD(1, 2, true, false);
 			} else {
 				int shape = e->crc_shape.load(std::memory_order_relaxed);
 				if (shape < 0 && tune && nblocks >= 4096) {

@@ -28,6 +28,8 @@ CFGS = [
     ("fold_pf_bv8", "fold", "1", "1", "0", "1", "0"),
     ("fold_pf_bv4", "fold", "1", "1", "0", "1", "4"),
     ("fold_bv4   ", "fold", "1", "1", "0", "0", "4"),
+    ("fold_n2_bv4", "fold", "1", "2", "0", "0", "4"),
+    ("fold_n4_bv4", "fold", "1", "4", "0", "0", "4"),
 ]
 
 
